@@ -1,0 +1,33 @@
+// Python bindings for the MI355X (gfx950) kernel extension.
+#include <torch/extension.h>
+#include <vector>
+
+// bn_act.hip
+std::vector<torch::Tensor> bn_act_fwd(torch::Tensor x, torch::Tensor gamma,
+                                      torch::Tensor beta, double eps,
+                                      double slope, bool act);
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
+                                      torch::Tensor mean, torch::Tensor rstd,
+                                      torch::Tensor gamma, torch::Tensor beta,
+                                      double slope, bool act);
+// pool.hip
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
+torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor mask, long H, long W);
+// softmax_ce.hip
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels);
+torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor labels, torch::Tensor gtask);
+// lslr.hip
+torch::Tensor lslr_fwd(torch::Tensor arena, torch::Tensor grad, torch::Tensor lr_vec);
+std::vector<torch::Tensor> lslr_bwd(torch::Tensor gout, torch::Tensor grad,
+                                    torch::Tensor lr_vec);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_act_fwd", &bn_act_fwd, "fused task-batched BN+leakyReLU fwd");
+  m.def("bn_act_bwd", &bn_act_bwd, "fused task-batched BN+leakyReLU bwd");
+  m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC maxpool 2x2 fwd");
+  m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC maxpool 2x2 bwd");
+  m.def("ce_fwd", &ce_fwd, "fused softmax-CE fwd");
+  m.def("ce_bwd", &ce_bwd, "fused softmax-CE bwd");
+  m.def("lslr_fwd", &lslr_fwd, "fused LSLR arena update fwd");
+  m.def("lslr_bwd", &lslr_bwd, "fused LSLR arena update bwd");
+}

@@ -1,0 +1,340 @@
+// Fused task-batched BatchNorm (batch statistics, training semantics) +
+// leaky-ReLU for NHWC-task-batched activations x[T, M, C] (M = NS*H*W).
+//
+// Replaces the reference's F.batch_norm + F.leaky_relu pair
+// (meta_neural_network_architectures.py:244-247, :383) — and, because the
+// whole task batch is processed at once, T x 2 ATen launches collapse into
+// 3 kernels (partial sums -> finalize -> normalize+act).
+//
+// Per-(task, channel) statistics: mean/var over the M image positions of
+// each task — matching the reference's per-task BN exactly.
+
+#include "common.h"
+
+using namespace maml355;
+
+// ---------------------------------------------------------------------------
+// Stage 1: partial sum / sum-of-squares accumulation into sums[T, 2, C].
+// Thread layout: lane -> channel (padded to 64), wave -> row group.
+// Coalesced: for a fixed row m, lanes 0..C-1 read consecutive addresses.
+// ---------------------------------------------------------------------------
+template <typename scalar_t>
+__global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
+                               float* __restrict__ sums,  // [T, 2, C]
+                               int T, long M, int C, int rows_per_block) {
+  const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
+  const int lanes_per_row = cpad;
+  const int rows_in_block = blockDim.x / lanes_per_row;   // e.g. 256/64 = 4
+  const int c = threadIdx.x % lanes_per_row;
+  const int rgroup = threadIdx.x / lanes_per_row;
+
+  const int t = blockIdx.x;                                // task
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  if (c >= C) return;
+
+  const scalar_t* xt = x + (long)t * M * C;
+  float s = 0.f, sq = 0.f;
+  const long row_end = min(row0 + rows_per_block, M);
+  for (long m = row0 + rgroup; m < row_end; m += rows_in_block) {
+    const float v = to_f32(xt[m * C + c]);
+    s += v;
+    sq += v * v;
+  }
+  // reduce the rows_in_block partial sums for channel c via LDS
+  extern __shared__ float lds[];  // [rows_in_block][cpad] x 2
+  float* lds_s = lds;
+  float* lds_q = lds + blockDim.x;
+  lds_s[threadIdx.x] = s;
+  lds_q[threadIdx.x] = sq;
+  __syncthreads();
+  if (rgroup == 0) {
+    for (int r = 1; r < rows_in_block; ++r) {
+      s += lds_s[r * lanes_per_row + c];
+      sq += lds_q[r * lanes_per_row + c];
+    }
+    atomicAdd(&sums[((long)t * 2 + 0) * C + c], s);
+    atomicAdd(&sums[((long)t * 2 + 1) * C + c], sq);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Stage 2: finalize mean / var / rstd per (t, c).
+// ---------------------------------------------------------------------------
+__global__ void bn_finalize_kernel(const float* __restrict__ sums, // [T,2,C]
+                                   float* __restrict__ mean,       // [T,C]
+                                   float* __restrict__ var,        // [T,C]
+                                   float* __restrict__ rstd,       // [T,C]
+                                   int T, long M, int C, float eps) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= T * C) return;
+  const int t = i / C, c = i % C;
+  const float m = sums[((long)t * 2 + 0) * C + c] / (float)M;
+  float v = sums[((long)t * 2 + 1) * C + c] / (float)M - m * m;
+  v = fmaxf(v, 0.f);
+  mean[i] = m;
+  var[i] = v;
+  rstd[i] = rsqrtf(v + eps);
+}
+
+// ---------------------------------------------------------------------------
+// Stage 3: normalize + affine + leaky-ReLU, elementwise (grid-stride).
+// gamma/beta are [C] (shared meta per-step params) or [T, C] (fast weights).
+// ---------------------------------------------------------------------------
+template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
+__global__ void bn_norm_act_kernel(const scalar_t* __restrict__ x,
+                                   scalar_t* __restrict__ y,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
+                                   int T, long M, int C, float slope) {
+  const long total = (long)T * M * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    const int c = (int)(i % C);
+    const int t = (int)(i / (M * (long)C));
+    const long tc = (long)t * C + c;
+    const float g = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
+    const float b = PER_TASK_AFFINE ? beta[tc] : beta[c];
+    float v = (to_f32(x[i]) - mean[tc]) * rstd[tc] * g + b;
+    if (ACT) v = v > 0.f ? v : v * slope;
+    y[i] = from_f32<scalar_t>(v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward stage 1: per-(t,c) sums  s1 = sum dy', s2 = sum dy' * xhat,
+// where dy' = dy * act'(pre-act) and xhat = (x - mean) * rstd.
+// ---------------------------------------------------------------------------
+template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
+__global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
+                                   const scalar_t* __restrict__ x,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
+                                   float* __restrict__ bsums,  // [T, 2, C]
+                                   int T, long M, int C, float slope,
+                                   int rows_per_block) {
+  const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
+  const int lanes_per_row = cpad;
+  const int rows_in_block = blockDim.x / lanes_per_row;
+  const int c = threadIdx.x % lanes_per_row;
+  const int rgroup = threadIdx.x / lanes_per_row;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  if (c >= C) return;
+
+  const long tc = (long)t * C + c;
+  const float mu = mean[tc], rs = rstd[tc];
+  const float g = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
+  const float b = PER_TASK_AFFINE ? beta[tc] : beta[c];
+  const scalar_t* xt = x + (long)t * M * C;
+  const scalar_t* dyt = dy + (long)t * M * C;
+  float s1 = 0.f, s2 = 0.f;
+  const long row_end = min(row0 + rows_per_block, M);
+  for (long m = row0 + rgroup; m < row_end; m += rows_in_block) {
+    const float xh = (to_f32(xt[m * C + c]) - mu) * rs;
+    float d = to_f32(dyt[m * C + c]);
+    if (ACT) {
+      const float pre = xh * g + b;
+      d *= (pre > 0.f) ? 1.f : slope;
+    }
+    s1 += d;
+    s2 += d * xh;
+  }
+  extern __shared__ float lds[];
+  float* lds_1 = lds;
+  float* lds_2 = lds + blockDim.x;
+  lds_1[threadIdx.x] = s1;
+  lds_2[threadIdx.x] = s2;
+  __syncthreads();
+  if (rgroup == 0) {
+    for (int r = 1; r < rows_in_block; ++r) {
+      s1 += lds_1[r * lanes_per_row + c];
+      s2 += lds_2[r * lanes_per_row + c];
+    }
+    atomicAdd(&bsums[((long)t * 2 + 0) * C + c], s1);
+    atomicAdd(&bsums[((long)t * 2 + 1) * C + c], s2);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward stage 2: dx = gamma*rstd * (dy' - s1/M - xhat * s2/M)
+// (dgamma = s2, dbeta = s1 are read from bsums by the wrapper.)
+// ---------------------------------------------------------------------------
+template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
+__global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
+                                 const scalar_t* __restrict__ x,
+                                 scalar_t* __restrict__ dx,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ beta,
+                                 const float* __restrict__ bsums,
+                                 int T, long M, int C, float slope) {
+  const long total = (long)T * M * C;
+  const float invM = 1.f / (float)M;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    const int c = (int)(i % C);
+    const int t = (int)(i / (M * (long)C));
+    const long tc = (long)t * C + c;
+    const float mu = mean[tc], rs = rstd[tc];
+    const float g = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
+    const float b = PER_TASK_AFFINE ? beta[tc] : beta[c];
+    const float xh = (to_f32(x[i]) - mu) * rs;
+    float d = to_f32(dy[i]);
+    if (ACT) {
+      const float pre = xh * g + b;
+      d *= (pre > 0.f) ? 1.f : slope;
+    }
+    const float s1 = bsums[((long)t * 2 + 0) * C + c];
+    const float s2 = bsums[((long)t * 2 + 1) * C + c];
+    dx[i] = from_f32<scalar_t>(g * rs * (d - s1 * invM - xh * s2 * invM));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C++ launchers (declared in bindings.cpp)
+// ---------------------------------------------------------------------------
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+namespace {
+
+constexpr int kRowsPerBlock = 256;
+constexpr int kThreads = 256;
+
+int norm_grid(long total) {
+  long blocks = (total + kThreads - 1) / kThreads;
+  return (int)std::min<long>(blocks, 4096);
+}
+
+template <typename scalar_t>
+void bn_fwd_impl(const torch::Tensor& x, const torch::Tensor& gamma,
+                 const torch::Tensor& beta, torch::Tensor& y,
+                 torch::Tensor& sums, torch::Tensor& mean, torch::Tensor& var,
+                 torch::Tensor& rstd, double eps, double slope, bool act,
+                 int T, long M, int C) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bool per_task = gamma.dim() == 2;
+  const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
+  dim3 sums_grid(T, (unsigned)((M + kRowsPerBlock - 1) / kRowsPerBlock));
+  const int threads = cpad * std::max<int>(1, kThreads / cpad);
+  const int lds_bytes = 2 * threads * sizeof(float);
+  hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), sums_grid, dim3(threads),
+                     lds_bytes, stream.stream(),
+                     reinterpret_cast<const scalar_t*>(x.data_ptr()), sums.data_ptr<float>(), T, M, C,
+                     kRowsPerBlock);
+  const int fin_threads = 256;
+  const int fin_blocks = (T * C + fin_threads - 1) / fin_threads;
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(fin_blocks), dim3(fin_threads), 0,
+                     stream.stream(), sums.data_ptr<float>(),
+                     mean.data_ptr<float>(), var.data_ptr<float>(),
+                     rstd.data_ptr<float>(), T, M, C, (float)eps);
+  const long total = (long)T * M * C;
+#define LAUNCH_NORM(PT, ACT_)                                                  \
+  hipLaunchKernelGGL((bn_norm_act_kernel<scalar_t, PT, ACT_>),                 \
+                     dim3(norm_grid(total)), dim3(kThreads), 0,                \
+                     stream.stream(), reinterpret_cast<const scalar_t*>(x.data_ptr()),                  \
+                     reinterpret_cast<scalar_t*>(y.data_ptr()), mean.data_ptr<float>(),           \
+                     rstd.data_ptr<float>(), gamma.data_ptr<float>(),          \
+                     beta.data_ptr<float>(), T, M, C, (float)slope)
+  if (per_task) { if (act) LAUNCH_NORM(true, true); else LAUNCH_NORM(true, false); }
+  else { if (act) LAUNCH_NORM(false, true); else LAUNCH_NORM(false, false); }
+#undef LAUNCH_NORM
+}
+
+template <typename scalar_t>
+void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& x,
+                 const torch::Tensor& mean, const torch::Tensor& rstd,
+                 const torch::Tensor& gamma, const torch::Tensor& beta,
+                 torch::Tensor& bsums, torch::Tensor& dx, double slope,
+                 bool act, int T, long M, int C) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bool per_task = gamma.dim() == 2;
+  const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
+  dim3 sums_grid(T, (unsigned)((M + kRowsPerBlock - 1) / kRowsPerBlock));
+  const int threads = cpad * std::max<int>(1, kThreads / cpad);
+  const int lds_bytes = 2 * threads * sizeof(float);
+  const long total = (long)T * M * C;
+#define LAUNCH_BWD(PT, ACT_)                                                   \
+  do {                                                                         \
+    hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t, PT, ACT_>), sums_grid,    \
+                       dim3(threads), lds_bytes, stream.stream(),              \
+                       reinterpret_cast<const scalar_t*>(dy.data_ptr()),       \
+                       reinterpret_cast<const scalar_t*>(x.data_ptr()),        \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),        \
+                       bsums.data_ptr<float>(), T, M, C, (float)slope,         \
+                       kRowsPerBlock);                                         \
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, PT, ACT_>),                 \
+                       dim3(norm_grid(total)), dim3(kThreads), 0,              \
+                       stream.stream(),                                        \
+                       reinterpret_cast<const scalar_t*>(dy.data_ptr()),       \
+                       reinterpret_cast<const scalar_t*>(x.data_ptr()),        \
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()),             \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),        \
+                       bsums.data_ptr<float>(), T, M, C, (float)slope);        \
+  } while (0)
+  if (per_task) { if (act) LAUNCH_BWD(true, true); else LAUNCH_BWD(true, false); }
+  else { if (act) LAUNCH_BWD(false, true); else LAUNCH_BWD(false, false); }
+#undef LAUNCH_BWD
+}
+
+}  // namespace
+
+// x: [T, M, C] contiguous (caller flattens NS*H*W -> M), fp32 or bf16.
+// gamma/beta: [C] or [T, C] fp32.
+std::vector<torch::Tensor> bn_act_fwd(torch::Tensor x, torch::Tensor gamma,
+                                      torch::Tensor beta, double eps,
+                                      double slope, bool act) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+  const int T = (int)x.size(0);
+  const long M = x.size(1);
+  const int C = (int)x.size(2);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sums = torch::zeros({T, 2, C}, opts);
+  auto mean = torch::empty({T, C}, opts);
+  auto var = torch::empty({T, C}, opts);
+  auto rstd = torch::empty({T, C}, opts);
+  auto y = torch::empty_like(x);
+  auto gc = gamma.contiguous().to(torch::kFloat32);
+  auto bc = beta.contiguous().to(torch::kFloat32);
+  if (x.scalar_type() == torch::kFloat32) {
+    bn_fwd_impl<float>(x, gc, bc, y, sums, mean, var, rstd, eps, slope, act, T, M, C);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    bn_fwd_impl<__hip_bfloat16>(x, gc, bc, y, sums, mean, var, rstd, eps, slope, act, T, M, C);
+  } else {
+    TORCH_CHECK(false, "bn_act_fwd: unsupported dtype");
+  }
+  return {y, mean, var, rstd};
+}
+
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
+                                      torch::Tensor mean, torch::Tensor rstd,
+                                      torch::Tensor gamma, torch::Tensor beta,
+                                      double slope, bool act) {
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 3);
+  auto dyc = dy.contiguous();
+  const int T = (int)x.size(0);
+  const long M = x.size(1);
+  const int C = (int)x.size(2);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto bsums = torch::zeros({T, 2, C}, opts);
+  auto dx = torch::empty_like(x);
+  auto gc = gamma.contiguous().to(torch::kFloat32);
+  auto bc = beta.contiguous().to(torch::kFloat32);
+  if (x.scalar_type() == torch::kFloat32) {
+    bn_bwd_impl<float>(dyc, x, mean, rstd, gc, bc, bsums, dx, slope, act, T, M, C);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    bn_bwd_impl<__hip_bfloat16>(dyc, x, mean, rstd, gc, bc, bsums, dx, slope, act, T, M, C);
+  } else {
+    TORCH_CHECK(false, "bn_act_bwd: unsupported dtype");
+  }
+  // dbeta = bsums[:,0,:], dgamma = bsums[:,1,:]  (per task; wrapper reduces
+  // over T when gamma is shared)
+  return {dx, bsums.select(1, 1).clone(), bsums.select(1, 0).clone()};
+}
