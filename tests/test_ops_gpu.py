@@ -88,16 +88,19 @@ def test_maxpool_fwd_bwd_double(H, W):
     y = ops.task_maxpool2x2(x)
     yr = ref.task_maxpool2x2(x.detach().cpu())
     torch.testing.assert_close(y.detach().cpu(), yr)
-    g = torch.randn_like(y)
+    # double-backward flows through grad_output only (the argmax mask is a
+    # discrete function of x), so g must require grad to exercise it
+    g = torch.randn_like(y).requires_grad_(True)
     (dx,) = torch.autograd.grad(y, x, g, create_graph=True)
     xr = x.detach().cpu().requires_grad_(True)
-    yr = ref.task_maxpool2x2(xr)
-    (dxr,) = torch.autograd.grad(yr, xr, g.cpu(), create_graph=True)
+    gr = g.detach().cpu().requires_grad_(True)
+    yr2 = ref.task_maxpool2x2(xr)
+    (dxr,) = torch.autograd.grad(yr2, xr, gr, create_graph=True)
     torch.testing.assert_close(dx.detach().cpu(), dxr.detach())
-    # double backward (gather path)
-    (ddx,) = torch.autograd.grad((dx ** 2).sum(), x)
-    (ddxr,) = torch.autograd.grad((dxr ** 2).sum(), xr)
-    torch.testing.assert_close(ddx.cpu(), ddxr, rtol=1e-4, atol=1e-5)
+    # double backward: d/dg of ||dx||^2  (gather path of _PoolBwdFn)
+    (ddg,) = torch.autograd.grad((dx ** 2).sum(), g)
+    (ddgr,) = torch.autograd.grad((dxr ** 2).sum(), gr)
+    torch.testing.assert_close(ddg.cpu(), ddgr, rtol=1e-4, atol=1e-5)
 
 
 @pytest.mark.parametrize("ways", [5, 20])
