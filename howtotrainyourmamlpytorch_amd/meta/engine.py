@@ -131,8 +131,14 @@ class MAMLFewShotClassifier(nn.Module):
         y_target = y_target.to(self.device, non_blocking=True)
 
         T = x_support.shape[0]
-        xs = x_support.reshape(T, -1, *x_support.shape[-3:]).float()
-        xt = x_target.reshape(T, -1, *x_target.shape[-3:]).float()
+        # bf16 activations on GPU (fp32 accumulate in the MFMA kernels and
+        # fp32 master weights in the arena); fp32 on CPU
+        act_dtype = torch.bfloat16 if (
+            self.device.type == "cuda"
+            and getattr(self.args, "compute_dtype", "bf16") == "bf16"
+        ) else torch.float32
+        xs = x_support.reshape(T, -1, *x_support.shape[-3:]).to(act_dtype)
+        xt = x_target.reshape(T, -1, *x_target.shape[-3:]).to(act_dtype)
         ys = y_support.reshape(T, -1).long()
         yt = y_target.reshape(T, -1).long()
 

@@ -20,6 +20,13 @@ torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor labels, torch::Tensor gt
 torch::Tensor lslr_fwd(torch::Tensor arena, torch::Tensor grad, torch::Tensor lr_vec);
 std::vector<torch::Tensor> lslr_bwd(torch::Tensor gout, torch::Tensor grad,
                                     torch::Tensor lr_vec);
+// tconv.hip
+torch::Tensor tconv_repack(torch::Tensor w, bool dgrad);
+torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
+                       c10::optional<torch::Tensor> bias, long pad,
+                       long Ho, long Wo);
+torch::Tensor tconv_wgrad(torch::Tensor dy, torch::Tensor x, long pad);
+std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd, "fused task-batched BN+leakyReLU fwd");
@@ -30,4 +37,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd, "fused softmax-CE bwd");
   m.def("lslr_fwd", &lslr_fwd, "fused LSLR arena update fwd");
   m.def("lslr_bwd", &lslr_bwd, "fused LSLR arena update bwd");
+  m.def("tconv_repack", &tconv_repack, "repack conv weights for MFMA (fwd/dgrad)");
+  m.def("tconv_mm", &tconv_mm, "task-batched MFMA 3x3 conv fwd/dgrad");
+  m.def("tconv_wgrad", &tconv_wgrad, "task-batched MFMA 3x3 conv wgrad");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
 }

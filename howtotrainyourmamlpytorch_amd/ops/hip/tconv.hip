@@ -1,0 +1,433 @@
+// Task-batched 3x3 convolution trio (fwd / dgrad / wgrad) as MFMA
+// implicit GEMM on gfx950 — the framework's flagship kernels.
+//
+// Replaces the reference's per-task F.conv2d calls
+// (meta_neural_network_architectures.py:89-97): each task in the meta-batch
+// has its OWN weights (fast weights), so ordinary batched conv does not
+// apply; MIOpen's grouped path falls back to naive NCHW kernels (measured
+// 20 ms/call wgrad — profiles/r01 baseline).  Here:
+//
+//   fwd   : Y[t]  = im2col(X[t])  @ Wp[t]        M=NS*Ho*Wo, N=Cout, K=9*Cin
+//   dgrad : dX[t] = im2col(dY[t]) @ Wp_flip[t]   (SAME kernel, weights
+//           repacked flipped+transposed — full correlation identity)
+//   wgrad : dW[t] = dY[t]^T @ im2col(X[t])       split-K, fp32 atomics
+//
+// Data: bf16 activations/weights (repacked from the fp32 arena), fp32
+// accumulate via v_mfma_f32_16x16x32_bf16.  Layout: NHWC (channel-
+// innermost = K-contiguous im2col rows).
+//
+// Geometry (fwd/dgrad): block = 256 threads = 4 waves; BM=64 output rows,
+// BN = Cout (<= 64, i.e. up to 4 n-tiles of 16), BK=32.  Wave w owns
+// 16-row m-subtile w and iterates all n-tiles.  A-tile staged by im2col
+// gather (LDS, +8 bf16 row pad), B-tile contiguous from the repacked
+// weights, transposed into LDS for K-contiguous fragment reads.
+
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+using namespace maml355;
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((__vector_size__(4 * sizeof(short)))) short short4v;
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
+
+// ---------------------------------------------------------------------------
+// Weight repack: W [T, F, C, 3, 3] fp32  ->  Wp [T, 9, Ci, Co] bf16
+//   fwd  : Wp[t][ky*3+kx][c][f] = W[t][f][c][ky][kx]          (Ci=C, Co=F)
+//   dgrad: Wp[t][ky*3+kx][f][c] = W[t][f][c][2-ky][2-kx]      (Ci=F, Co=C)
+// ---------------------------------------------------------------------------
+__global__ void repack_weight_kernel(const float* __restrict__ w,
+                                     bf16* __restrict__ wp,
+                                     int T, int F, int C, bool dgrad) {
+  const long total = (long)T * F * C * 9;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    long r = i;
+    const int kx = (int)(r % 3); r /= 3;
+    const int ky = (int)(r % 3); r /= 3;
+    const int c = (int)(r % C); r /= C;
+    const int f = (int)(r % F); r /= F;
+    const long t = r;
+    const float v = w[i];
+    long o;
+    if (!dgrad) {
+      o = (((t * 9 + (long)(ky * 3 + kx)) * C + c) * F) + f;
+    } else {
+      o = (((t * 9 + (long)((2 - ky) * 3 + (2 - kx))) * F + f) * C) + c;
+    }
+    wp[o] = __float2bfloat16(v);
+  }
+}
+
+// fp32 wgrad accumulator [T, 9C, F] -> dW [T, F, C, 3, 3] fp32
+__global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
+                                      float* __restrict__ dw,
+                                      int T, int F, int C) {
+  const long total = (long)T * F * C * 9;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    long r = i;
+    const int kx = (int)(r % 3); r /= 3;
+    const int ky = (int)(r % 3); r /= 3;
+    const int c = (int)(r % C); r /= C;
+    const int f = (int)(r % F); r /= F;
+    const long t = r;
+    dw[i] = acc[((t * 9 + (long)(ky * 3 + kx)) * C + c) * F + f];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fwd / dgrad kernel.
+//   X  [T, NB, H, W, Ci]   bf16 (NB = images per task)
+//   Wp [T, 9, Ci, Co]      bf16 (repacked; flipped/transposed for dgrad)
+//   bias [T, Co] fp32 or nullptr
+//   Y  [T, NB, Ho, Wo, Co] bf16
+// pad is the im2col pad (fwd: p; dgrad: 2-p).  stride==1 only (stride-2
+// configs use the ATen fallback path).
+// ---------------------------------------------------------------------------
+#define BM 64
+#define BK 32
+#define APAD 8   // bf16 row pad for LDS bank spread
+
+__global__ __launch_bounds__(256, 2)
+void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
+                     const float* __restrict__ bias, bf16* __restrict__ Y,
+                     int T, int NB, int H, int W, int Ci,
+                     int Ho, int Wo, int Co, int pad) {
+  const int t = blockIdx.y;
+  const long Mtot = (long)NB * Ho * Wo;
+  const long m0 = (long)blockIdx.x * BM;
+  const int K9 = 9 * Ci;
+  const int ksteps = (K9 + BK - 1) / BK;
+  const int ntiles = (Co + 15) / 16;
+
+  __shared__ short lds_a[BM][BK + APAD];
+  __shared__ short lds_bt[64][BK + APAD];   // B^T: [col][k]
+  __shared__ int row_h[BM], row_w[BM], row_n[BM];
+  __shared__ int ktab_dy[BK], ktab_dx[BK], ktab_c[BK];
+
+  // per-block row decode (once)
+  for (int m = threadIdx.x; m < BM; m += blockDim.x) {
+    const long mg = m0 + m;
+    if (mg < Mtot) {
+      const int wo = (int)(mg % Wo);
+      const int ho = (int)((mg / Wo) % Ho);
+      const int n = (int)(mg / ((long)Wo * Ho));
+      row_h[m] = ho; row_w[m] = wo; row_n[m] = n;
+    } else {
+      row_n[m] = -1;
+    }
+  }
+
+  const bf16* Xt = X + (long)t * NB * H * W * Ci;
+  const bf16* Wt = Wp + (long)t * 9 * Ci * Co;
+
+  const int wave = threadIdx.x / WAVE;          // 0..3: m-subtile
+  const int lane = threadIdx.x % WAVE;
+  const int fr = lane & 15;                      // fragment row/col
+  const int fk = lane >> 4;                      // k-group 0..3
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int ks = 0; ks < ksteps; ++ks) {
+    const int k0 = ks * BK;
+    // k-table for this step
+    __syncthreads();
+    if (threadIdx.x < BK) {
+      const int k = k0 + (int)threadIdx.x;
+      if (k < K9) {
+        const int kyx = k / Ci;
+        ktab_c[threadIdx.x] = k - kyx * Ci;
+        ktab_dy[threadIdx.x] = kyx / 3;
+        ktab_dx[threadIdx.x] = kyx % 3;
+      } else {
+        ktab_c[threadIdx.x] = -1;
+      }
+    }
+    __syncthreads();
+
+    // stage A: BM x BK im2col gather
+    for (int e = threadIdx.x; e < BM * BK; e += blockDim.x) {
+      const int m = e >> 5;          // BK == 32
+      const int kk = e & 31;
+      short v = 0;
+      const int n = row_n[m];
+      const int c = ktab_c[kk];
+      if (n >= 0 && c >= 0) {
+        const int h = row_h[m] + ktab_dy[kk] - pad;
+        const int w = row_w[m] + ktab_dx[kk] - pad;
+        if (h >= 0 && h < H && w >= 0 && w < W) {
+          v = ((const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c];
+        }
+      }
+      lds_a[m][kk] = v;
+    }
+    // stage B^T: from Wp contiguous [BK][Co] block at linear k0*Co
+    for (int e = threadIdx.x; e < BK * Co; e += blockDim.x) {
+      const int f = e % Co;
+      const int kk = e / Co;
+      short v = 0;
+      if (k0 + kk < K9) v = ((const short*)Wt)[(long)(k0 + kk) * Co + f];
+      lds_bt[f][kk] = v;
+    }
+    __syncthreads();
+
+    // fragments + MFMA
+    bf16x8 a = *(const bf16x8*)&lds_a[wave * 16 + fr][fk * 8];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      if (nt < ntiles) {
+        bf16x8 b = *(const bf16x8*)&lds_bt[nt * 16 + fr][fk * 8];
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + j
+  bf16* Yt = Y + (long)t * Mtot * Co;
+#pragma unroll
+  for (int nt = 0; nt < 4; ++nt) {
+    if (nt >= ntiles) break;
+    const int col = nt * 16 + fr;
+    if (col >= Co) continue;
+    const float bv = bias ? bias[(long)t * Co + col] : 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int row = wave * 16 + fk * 4 + j;
+      const long mg = m0 + row;
+      if (mg < Mtot) {
+        ((short*)Yt)[mg * Co + col] =
+            (short)__bfloat16_as_short(__float2bfloat16(acc[nt][j] + bv));
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad kernel: dWacc[t][n=9C][f] += sum_k dY[t,k,f] * im2col(X)[t,k,n]
+// Block: 256 thr = 4 waves; wave w owns n-subtile w (16 cols), iterates
+// m-tiles over F.  K-chunked grid with fp32 atomicAdd.
+//   dY [T, NB, Ho, Wo, F] bf16 ; X [T, NB, H, W, C] bf16
+// ---------------------------------------------------------------------------
+#define WG_KCHUNK 4096
+
+__global__ __launch_bounds__(256, 2)
+void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
+                        float* __restrict__ dWacc,
+                        int T, int NB, int H, int W, int C,
+                        int Ho, int Wo, int F, int pad) {
+  const int t = blockIdx.z;
+  const int n0 = blockIdx.x * 64;          // column block within 9C
+  const int N9 = 9 * C;
+  const long Ktot = (long)NB * Ho * Wo;
+  const long kchunk0 = (long)blockIdx.y * WG_KCHUNK;
+  const long kchunk_end = min(kchunk0 + (long)WG_KCHUNK, Ktot);
+  const int mtiles = (F + 15) / 16;
+
+  __shared__ short lds_at[64][BK + APAD];   // dY^T tile: [f][k]
+  __shared__ short lds_bt[64][BK + APAD];   // im2col^T tile: [n][k]
+  __shared__ int ntab_dy[64], ntab_dx[64], ntab_c[64];
+
+  // n-table (once): n -> (dy, dx, c)
+  for (int e = threadIdx.x; e < 64; e += blockDim.x) {
+    const int n = n0 + e;
+    if (n < N9) {
+      const int kyx = n / C;
+      ntab_c[e] = n - kyx * C;
+      ntab_dy[e] = kyx / 3;
+      ntab_dx[e] = kyx % 3;
+    } else {
+      ntab_c[e] = -1;
+    }
+  }
+
+  const bf16* dYt = dY + (long)t * Ktot * F;
+  const bf16* Xt = X + (long)t * NB * H * W * C;
+
+  const int wave = threadIdx.x / WAVE;      // n-subtile
+  const int lane = threadIdx.x % WAVE;
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long k0 = kchunk0; k0 < kchunk_end; k0 += BK) {
+    __syncthreads();
+    // stage dY^T: element (f, kk) <- dY[k0+kk][f]
+    for (int e = threadIdx.x; e < BK * F; e += blockDim.x) {
+      const int f = e % F;
+      const int kk = e / F;
+      short v = 0;
+      if (k0 + kk < kchunk_end) v = ((const short*)dYt)[(k0 + kk) * F + f];
+      lds_at[f][kk] = v;
+    }
+    // stage im2col^T: element (n, kk) <- X[inpos(k0+kk, n)]
+    for (int e = threadIdx.x; e < BK * 64; e += blockDim.x) {
+      const int ncol = e & 63;
+      const int kk = e >> 6;
+      short v = 0;
+      const long k = k0 + kk;
+      const int c = ntab_c[ncol];
+      if (k < kchunk_end && c >= 0) {
+        const int wo = (int)(k % Wo);
+        const int ho = (int)((k / Wo) % Ho);
+        const int n = (int)(k / ((long)Wo * Ho));
+        const int h = ho + ntab_dy[ncol] - pad;
+        const int w = wo + ntab_dx[ncol] - pad;
+        if (h >= 0 && h < H && w >= 0 && w < W) {
+          v = ((const short*)Xt)[(((long)n * H + h) * W + w) * C + c];
+        }
+      }
+      lds_bt[ncol][kk] = v;
+    }
+    __syncthreads();
+
+    bf16x8 b = *(const bf16x8*)&lds_bt[wave * 16 + fr][fk * 8];
+#pragma unroll
+    for (int mt = 0; mt < 4; ++mt) {
+      if (mt < mtiles) {
+        bf16x8 a = *(const bf16x8*)&lds_at[mt * 16 + fr][fk * 8];
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
+      }
+    }
+  }
+
+  // accumulate: C/D col = lane&15 (n within wave tile), row = fk*4+j (f)
+  float* dWt = dWacc + (long)t * N9 * F;
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+    if (mt >= mtiles) break;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int f = mt * 16 + fk * 4 + j;
+      const int n = n0 + wave * 16 + fr;
+      if (f < F && n < N9) {
+        atomicAdd(&dWt[(long)n * F + f], acc[mt][j]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe: D[16x16] = A[16x32] x B[32x16], single wave.
+// Used by the GPU tests to pin the fragment layout against torch.matmul
+// with asymmetric random matrices (guide rule G9).
+// ---------------------------------------------------------------------------
+__global__ void mfma_probe_kernel(const bf16* __restrict__ A,
+                                  const bf16* __restrict__ B,
+                                  float* __restrict__ D) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = ((const short*)A)[fr * 32 + fk * 8 + j];        // A[row][k]
+    b[j] = ((const short*)B)[(fk * 8 + j) * 16 + fr];      // B[k][col]
+  }
+  f32x4 d = {0.f, 0.f, 0.f, 0.f};
+  d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, d, 0, 0, 0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    D[(fk * 4 + j) * 16 + fr] = d[j];                      // D[row][col]
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+namespace {
+int ew_grid2(long total, int threads) {
+  long blocks = (total + threads - 1) / threads;
+  return (int)std::min<long>(blocks, 4096);
+}
+}  // namespace
+
+torch::Tensor tconv_repack(torch::Tensor w, bool dgrad) {
+  TORCH_CHECK(w.is_cuda() && w.dim() == 5 && w.size(3) == 3 && w.size(4) == 3);
+  auto wc = w.contiguous().to(torch::kFloat32);
+  const int T = (int)w.size(0), F = (int)w.size(1), C = (int)w.size(2);
+  auto wp = torch::empty({T, 9, dgrad ? F : C, dgrad ? C : F},
+                         w.options().dtype(torch::kBFloat16));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const long total = (long)T * F * C * 9;
+  hipLaunchKernelGGL(repack_weight_kernel, dim3(ew_grid2(total, 256)), dim3(256),
+                     0, stream.stream(), wc.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(wp.data_ptr()), T, F, C, dgrad);
+  return wp;
+}
+
+// x [T, NB, H, W, Ci] bf16 ; wp [T, 9, Ci, Co] bf16 ; bias [T, Co] fp32 / undef
+torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
+                       c10::optional<torch::Tensor> bias, long pad,
+                       long Ho, long Wo) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 5 && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "tconv_mm needs bf16");
+  const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3), Ci = (int)x.size(4);
+  const int Co = (int)wp.size(3);
+  TORCH_CHECK(wp.size(2) == Ci && Co <= 64, "Co must be <= 64");
+  auto y = torch::empty({T, NB, Ho, Wo, Co}, x.options());
+  const long Mtot = (long)NB * Ho * Wo;
+  dim3 grid((unsigned)((Mtot + BM - 1) / BM), T);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const float* bptr = nullptr;
+  torch::Tensor bc;
+  if (bias.has_value()) {
+    bc = bias->contiguous().to(torch::kFloat32);
+    bptr = bc.data_ptr<float>();
+  }
+  hipLaunchKernelGGL(tconv_mm_kernel, grid, dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const bf16*>(x.data_ptr()),
+                     reinterpret_cast<const bf16*>(wp.data_ptr()), bptr,
+                     reinterpret_cast<bf16*>(y.data_ptr()),
+                     T, NB, H, W, Ci, (int)Ho, (int)Wo, Co, (int)pad);
+  return y;
+}
+
+// dy [T, NB, Ho, Wo, F] bf16 ; x [T, NB, H, W, C] bf16 -> dw [T, F, C, 3, 3] fp32
+torch::Tensor tconv_wgrad(torch::Tensor dy, torch::Tensor x, long pad) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda());
+  auto dyc = dy.contiguous();
+  auto xc = x.contiguous();
+  const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3), C = (int)x.size(4);
+  const int Ho = (int)dy.size(2), Wo = (int)dy.size(3), F = (int)dy.size(4);
+  TORCH_CHECK(F <= 64, "F must be <= 64");
+  const int N9 = 9 * C;
+  auto acc = torch::zeros({T, N9, F}, x.options().dtype(torch::kFloat32));
+  const long Ktot = (long)NB * Ho * Wo;
+  dim3 grid((unsigned)((N9 + 63) / 64),
+            (unsigned)((Ktot + WG_KCHUNK - 1) / WG_KCHUNK), T);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const bf16*>(dyc.data_ptr()),
+                     reinterpret_cast<const bf16*>(xc.data_ptr()),
+                     acc.data_ptr<float>(), T, NB, H, W, C, Ho, Wo, F, (int)pad);
+  auto dw = torch::empty({T, F, C, 3, 3}, x.options().dtype(torch::kFloat32));
+  const long total = (long)T * F * C * 9;
+  hipLaunchKernelGGL(wgrad_finalize_kernel, dim3(ew_grid2(total, 256)),
+                     dim3(256), 0, stream.stream(), acc.data_ptr<float>(),
+                     dw.data_ptr<float>(), T, F, C);
+  return dw;
+}
+
+std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
+  auto Ac = A.contiguous().to(torch::kBFloat16);
+  auto Bc = B.contiguous().to(torch::kBFloat16);
+  auto D = torch::zeros({16, 16}, A.options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream.stream(),
+                     reinterpret_cast<const bf16*>(Ac.data_ptr()),
+                     reinterpret_cast<const bf16*>(Bc.data_ptr()),
+                     D.data_ptr<float>());
+  return {D};
+}
