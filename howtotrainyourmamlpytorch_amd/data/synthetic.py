@@ -12,7 +12,16 @@ import torch
 
 
 class SyntheticEpisodeStream:
-    def __init__(self, args, rank: int = 0, world_size: int = 1):
+    """``structured=False`` (default): pure random pixels — right shapes and
+    bandwidth for benchmarking.  ``structured=True``: a *learnable* few-shot
+    task distribution — ``num_latent_classes`` fixed random prototypes;
+    episodes sample N of them and emit prototype+noise images with
+    consistent labels, so meta-training must raise target accuracy above
+    chance (used by the GPU learning test)."""
+
+    def __init__(self, args, rank: int = 0, world_size: int = 1,
+                 structured: bool = False, num_latent_classes: int = 64,
+                 noise: float = 0.35):
         self.args = args
         self.rank = rank
         self.world_size = world_size
@@ -21,6 +30,12 @@ class SyntheticEpisodeStream:
             raise ValueError("batch_size must divide world_size")
         self.local_batch = self.global_batch // world_size
         self.total_train_iters_produced = 0
+        self.structured = structured
+        self.noise = noise
+        if structured:
+            g = torch.Generator().manual_seed(1234 + args.seed)
+            c, h, w = args.image_channels, args.image_height, args.image_width
+            self.prototypes = torch.randn(num_latent_classes, c, h, w, generator=g)
 
     def continue_from_iter(self, current_iter: int) -> None:
         self.total_train_iters_produced = current_iter
@@ -30,8 +45,16 @@ class SyntheticEpisodeStream:
         g = torch.Generator().manual_seed(seed & 0x7FFFFFFF)
         n, s, t = a.num_classes_per_set, a.num_samples_per_class, a.num_target_samples
         c, h, w = a.image_channels, a.image_height, a.image_width
-        xs = torch.rand(n, s, c, h, w, generator=g)
-        xt = torch.rand(n, t, c, h, w, generator=g)
+        if self.structured:
+            cls = torch.randperm(self.prototypes.shape[0], generator=g)[:n]
+            proto = self.prototypes[cls]                       # [n, c, h, w]
+            xs = proto.unsqueeze(1) + self.noise * torch.randn(
+                n, s, c, h, w, generator=g)
+            xt = proto.unsqueeze(1) + self.noise * torch.randn(
+                n, t, c, h, w, generator=g)
+        else:
+            xs = torch.rand(n, s, c, h, w, generator=g)
+            xt = torch.rand(n, t, c, h, w, generator=g)
         ys = torch.arange(n).view(n, 1).expand(n, s).contiguous()
         yt = torch.arange(n).view(n, 1).expand(n, t).contiguous()
         return xs, xt, ys, yt
