@@ -88,7 +88,7 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
 // configs use the ATen fallback path).
 // ---------------------------------------------------------------------------
 #define BM 64
-#define BK 32
+#define BK 64    // K-step: 2 MFMA K-slices per barrier
 #define APAD 8   // bf16 row pad for LDS bank spread
 
 __global__ __launch_bounds__(256, 2)
@@ -150,10 +150,10 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     }
     __syncthreads();
 
-    // stage A: BM x BK im2col gather
+    // stage A: BM x BK im2col gather; e -> (m = e>>6, kk = e&63)
     for (int e = threadIdx.x; e < BM * BK; e += blockDim.x) {
-      const int m = e >> 5;          // BK == 32
-      const int kk = e & 31;
+      const int m = e >> 6;          // BK == 64
+      const int kk = e & 63;
       short v = 0;
       const int n = row_n[m];
       const int c = ktab_c[kk];
@@ -166,23 +166,27 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
       }
       lds_a[m][kk] = v;
     }
-    // stage B^T: from Wp contiguous [BK][Co] block at linear k0*Co
-    for (int e = threadIdx.x; e < BK * Co; e += blockDim.x) {
-      const int f = e % Co;
-      const int kk = e / Co;
+    // stage B^T: from Wp contiguous [BK][Co] block at linear k0*Co;
+    // e -> (f = e&63, kk = e>>6) avoids runtime division
+    for (int e = threadIdx.x; e < BK * 64; e += blockDim.x) {
+      const int f = e & 63;
+      const int kk = e >> 6;
       short v = 0;
-      if (k0 + kk < K9) v = ((const short*)Wt)[(long)(k0 + kk) * Co + f];
+      if (f < Co && k0 + kk < K9) v = ((const short*)Wt)[(long)(k0 + kk) * Co + f];
       lds_bt[f][kk] = v;
     }
     __syncthreads();
 
-    // fragments + MFMA
-    bf16x8 a = *(const bf16x8*)&lds_a[wave * 16 + fr][fk * 8];
+    // fragments + MFMA (2 K-slices per staged tile)
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
-      if (nt < ntiles) {
-        bf16x8 b = *(const bf16x8*)&lds_bt[nt * 16 + fr][fk * 8];
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8 a = *(const bf16x8*)&lds_a[wave * 16 + fr][ks * 32 + fk * 8];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        if (nt < ntiles) {
+          bf16x8 b = *(const bf16x8*)&lds_bt[nt * 16 + fr][ks * 32 + fk * 8];
+          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+        }
       }
     }
   }
@@ -214,6 +218,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 //   dY [T, NB, Ho, Wo, F] bf16 ; X [T, NB, H, W, C] bf16
 // ---------------------------------------------------------------------------
 #define WG_KCHUNK 4096
+#define WBK 64   // wgrad K-step (2 MFMA K-slices per barrier)
 
 __global__ __launch_bounds__(256, 2)
 void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
@@ -228,9 +233,10 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   const long kchunk_end = min(kchunk0 + (long)WG_KCHUNK, Ktot);
   const int mtiles = (F + 15) / 16;
 
-  __shared__ short lds_at[64][BK + APAD];   // dY^T tile: [f][k]
-  __shared__ short lds_bt[64][BK + APAD];   // im2col^T tile: [n][k]
+  __shared__ short lds_at[64][WBK + APAD];  // dY^T tile: [f][k]
+  __shared__ short lds_bt[64][WBK + APAD];  // im2col^T tile: [n][k]
   __shared__ int ntab_dy[64], ntab_dx[64], ntab_c[64];
+  __shared__ int ktab_n[WBK], ktab_h[WBK], ktab_w[WBK];  // k -> image pos
 
   // n-table (once): n -> (dy, dx, c)
   for (int e = threadIdx.x; e < 64; e += blockDim.x) {
@@ -257,43 +263,57 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
 #pragma unroll
   for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 
-  for (long k0 = kchunk0; k0 < kchunk_end; k0 += BK) {
+  for (long k0 = kchunk0; k0 < kchunk_end; k0 += WBK) {
     __syncthreads();
-    // stage dY^T: element (f, kk) <- dY[k0+kk][f]
-    for (int e = threadIdx.x; e < BK * F; e += blockDim.x) {
-      const int f = e % F;
-      const int kk = e / F;
+    // k-position table: one decode per k instead of per staged element
+    if (threadIdx.x < WBK) {
+      const long k = k0 + threadIdx.x;
+      if (k < kchunk_end) {
+        const int wo = (int)(k % Wo);
+        const long r = k / Wo;
+        ktab_w[threadIdx.x] = wo;
+        ktab_h[threadIdx.x] = (int)(r % Ho);
+        ktab_n[threadIdx.x] = (int)(r / Ho);
+      } else {
+        ktab_n[threadIdx.x] = -1;
+      }
+    }
+    __syncthreads();
+    // stage dY^T: element (f, kk) <- dY[k0+kk][f]; e -> (f = e&63, kk = e>>6)
+    for (int e = threadIdx.x; e < WBK * 64; e += blockDim.x) {
+      const int f = e & 63;
+      const int kk = e >> 6;
       short v = 0;
-      if (k0 + kk < kchunk_end) v = ((const short*)dYt)[(k0 + kk) * F + f];
+      if (f < F && ktab_n[kk] >= 0) v = ((const short*)dYt)[(k0 + kk) * F + f];
       lds_at[f][kk] = v;
     }
     // stage im2col^T: element (n, kk) <- X[inpos(k0+kk, n)]
-    for (int e = threadIdx.x; e < BK * 64; e += blockDim.x) {
+    for (int e = threadIdx.x; e < WBK * 64; e += blockDim.x) {
       const int ncol = e & 63;
       const int kk = e >> 6;
       short v = 0;
-      const long k = k0 + kk;
+      const int nimg = ktab_n[kk];
       const int c = ntab_c[ncol];
-      if (k < kchunk_end && c >= 0) {
-        const int wo = (int)(k % Wo);
-        const int ho = (int)((k / Wo) % Ho);
-        const int n = (int)(k / ((long)Wo * Ho));
-        const int h = ho + ntab_dy[ncol] - pad;
-        const int w = wo + ntab_dx[ncol] - pad;
+      if (nimg >= 0 && c >= 0) {
+        const int h = ktab_h[kk] + ntab_dy[ncol] - pad;
+        const int w = ktab_w[kk] + ntab_dx[ncol] - pad;
         if (h >= 0 && h < H && w >= 0 && w < W) {
-          v = ((const short*)Xt)[(((long)n * H + h) * W + w) * C + c];
+          v = ((const short*)Xt)[(((long)nimg * H + h) * W + w) * C + c];
         }
       }
       lds_bt[ncol][kk] = v;
     }
     __syncthreads();
 
-    bf16x8 b = *(const bf16x8*)&lds_bt[wave * 16 + fr][fk * 8];
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
-      if (mt < mtiles) {
-        bf16x8 a = *(const bf16x8*)&lds_at[mt * 16 + fr][fk * 8];
-        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
+    for (int ks = 0; ks < WBK / 32; ++ks) {
+      bf16x8 b = *(const bf16x8*)&lds_bt[wave * 16 + fr][ks * 32 + fk * 8];
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        if (mt < mtiles) {
+          bf16x8 a = *(const bf16x8*)&lds_at[mt * 16 + fr][ks * 32 + fk * 8];
+          acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
+        }
       }
     }
   }
