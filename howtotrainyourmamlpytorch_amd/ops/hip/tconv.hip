@@ -150,30 +150,63 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     }
     __syncthreads();
 
-    // stage A: BM x BK im2col gather; e -> (m = e>>6, kk = e&63)
-    for (int e = threadIdx.x; e < BM * BK; e += blockDim.x) {
-      const int m = e >> 6;          // BK == 64
-      const int kk = e & 63;
-      short v = 0;
+    // stage A (im2col): slot s -> (m = s>>3, kk0 = (s&7)*8); vector path
+    // when the 8-k run stays inside one (ky,kx) slice (contiguous c).
+    for (int s = threadIdx.x; s < BM * 8; s += blockDim.x) {
+      const int m = s >> 3;
+      const int kk0 = (s & 7) * 8;
       const int n = row_n[m];
-      const int c = ktab_c[kk];
-      if (n >= 0 && c >= 0) {
-        const int h = row_h[m] + ktab_dy[kk] - pad;
-        const int w = row_w[m] + ktab_dx[kk] - pad;
-        if (h >= 0 && h < H && w >= 0 && w < W) {
-          v = ((const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c];
+      const int c0 = ktab_c[kk0];
+      const int c7 = ktab_c[kk0 + 7];
+      if ((Ci % 8 == 0) && c0 >= 0 && c7 == c0 + 7) {
+        bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (n >= 0) {
+          const int h = row_h[m] + ktab_dy[kk0] - pad;
+          const int w = row_w[m] + ktab_dx[kk0] - pad;
+          if (h >= 0 && h < H && w >= 0 && w < W) {
+            v = *(const bf16x8*)&(
+                (const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c0];
+          }
+        }
+        *(bf16x8*)&lds_a[m][kk0] = v;
+      } else {
+        for (int j = 0; j < 8; ++j) {
+          const int kk = kk0 + j;
+          const int c = ktab_c[kk];
+          short v = 0;
+          if (n >= 0 && c >= 0) {
+            const int h = row_h[m] + ktab_dy[kk] - pad;
+            const int w = row_w[m] + ktab_dx[kk] - pad;
+            if (h >= 0 && h < H && w >= 0 && w < W) {
+              v = ((const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c];
+            }
+          }
+          lds_a[m][kk] = v;
         }
       }
-      lds_a[m][kk] = v;
     }
-    // stage B^T: from Wp contiguous [BK][Co] block at linear k0*Co;
-    // e -> (f = e&63, kk = e>>6) avoids runtime division
-    for (int e = threadIdx.x; e < BK * 64; e += blockDim.x) {
-      const int f = e & 63;
-      const int kk = e >> 6;
-      short v = 0;
-      if (f < Co && k0 + kk < K9) v = ((const short*)Wt)[(long)(k0 + kk) * Co + f];
-      lds_bt[f][kk] = v;
+    // stage B^T from the contiguous repacked block at linear k0*Co:
+    // slot s -> (kk = s>>3, f0 = (s&7)*8), bf16x8 load + transposed writes
+    for (int s = threadIdx.x; s < BK * 8; s += blockDim.x) {
+      const int kk = s >> 3;
+      const int f0 = (s & 7) * 8;
+      if ((Co % 8 == 0) && f0 + 8 <= Co) {
+        bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (k0 + kk < K9) {
+          v = *(const bf16x8*)&((const short*)Wt)[(long)(k0 + kk) * Co + f0];
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) lds_bt[f0 + j][kk] = v[j];
+      } else {
+        for (int j = 0; j < 8 && f0 + j < 64; ++j) {
+          const int f = f0 + j;
+          short v = 0;
+          if (f < Co && k0 + kk < K9) {
+            v = ((const short*)Wt)[(long)(k0 + kk) * Co + f];
+          }
+          lds_bt[f][kk] = v;
+        }
+      }
     }
     __syncthreads();
 
@@ -279,29 +312,58 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
       }
     }
     __syncthreads();
-    // stage dY^T: element (f, kk) <- dY[k0+kk][f]; e -> (f = e&63, kk = e>>6)
-    for (int e = threadIdx.x; e < WBK * 64; e += blockDim.x) {
-      const int f = e & 63;
-      const int kk = e >> 6;
-      short v = 0;
-      if (f < F && ktab_n[kk] >= 0) v = ((const short*)dYt)[(k0 + kk) * F + f];
-      lds_at[f][kk] = v;
+    // stage dY^T, vectorized: slot s -> (kk = s>>3, f0 = (s&7)*8); one
+    // bf16x8 load of 8 consecutive f, 8 scalar LDS writes (transpose).
+    // F is always a multiple of 16 here (conv out-channels 48/64), and
+    // rows beyond F are never read by the fragments (mtiles*16 == F).
+    for (int s = threadIdx.x; s < WBK * 8; s += blockDim.x) {
+      const int kk = s >> 3;
+      const int f0 = (s & 7) * 8;
+      if (f0 >= F) continue;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (ktab_n[kk] >= 0) {
+        v = *(const bf16x8*)&((const short*)dYt)[(k0 + kk) * F + f0];
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_at[f0 + j][kk] = v[j];
     }
-    // stage im2col^T: element (n, kk) <- X[inpos(k0+kk, n)]
-    for (int e = threadIdx.x; e < WBK * 64; e += blockDim.x) {
-      const int ncol = e & 63;
-      const int kk = e >> 6;
-      short v = 0;
+    // stage im2col^T: vector fast path when the 8-column run stays inside
+    // one (ky,kx) slice (contiguous c, 16B-aligned when C % 8 == 0);
+    // scalar fallback otherwise (first-layer C in {1,3}).
+    for (int s = threadIdx.x; s < WBK * 8; s += blockDim.x) {
+      const int kk = s >> 3;
+      const int n8 = (s & 7) * 8;
       const int nimg = ktab_n[kk];
-      const int c = ntab_c[ncol];
-      if (nimg >= 0 && c >= 0) {
-        const int h = ktab_h[kk] + ntab_dy[ncol] - pad;
-        const int w = ktab_w[kk] + ntab_dx[ncol] - pad;
-        if (h >= 0 && h < H && w >= 0 && w < W) {
-          v = ((const short*)Xt)[(((long)nimg * H + h) * W + w) * C + c];
+      const int c0 = ntab_c[n8];
+      const int c7 = ntab_c[n8 + 7];
+      const bool vec_ok = (C % 8 == 0) && c0 >= 0 && c7 == c0 + 7;
+      if (vec_ok) {
+        bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (nimg >= 0) {
+          const int h = ktab_h[kk] + ntab_dy[n8] - pad;
+          const int w = ktab_w[kk] + ntab_dx[n8] - pad;
+          if (h >= 0 && h < H && w >= 0 && w < W) {
+            v = *(const bf16x8*)&(
+                (const short*)Xt)[(((long)nimg * H + h) * W + w) * C + c0];
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) lds_bt[n8 + j][kk] = v[j];
+      } else {
+        for (int j = 0; j < 8; ++j) {
+          const int ncol = n8 + j;
+          const int c = ntab_c[ncol];
+          short v = 0;
+          if (nimg >= 0 && c >= 0) {
+            const int h = ktab_h[kk] + ntab_dy[ncol] - pad;
+            const int w = ktab_w[kk] + ntab_dx[ncol] - pad;
+            if (h >= 0 && h < H && w >= 0 && w < W) {
+              v = ((const short*)Xt)[(((long)nimg * H + h) * W + w) * C + c];
+            }
+          }
+          lds_bt[ncol][kk] = v;
         }
       }
-      lds_bt[ncol][kk] = v;
     }
     __syncthreads();
 
