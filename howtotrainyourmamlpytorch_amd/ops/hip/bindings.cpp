@@ -16,6 +16,14 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor mask, long H, long 
 // softmax_ce.hip
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels);
 torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor labels, torch::Tensor gtask);
+std::vector<torch::Tensor> ce_dbwd(torch::Tensor probs, torch::Tensor labels,
+                                   torch::Tensor gdl, torch::Tensor gtask);
+// bn_dbwd.hip
+std::vector<torch::Tensor> bn_act_dbwd(torch::Tensor x, torch::Tensor u,
+                                       torch::Tensor gx, torch::Tensor mean,
+                                       torch::Tensor rstd, torch::Tensor gamma,
+                                       torch::Tensor beta, torch::Tensor ggam,
+                                       torch::Tensor gbet, double slope, bool act);
 // lslr.hip
 torch::Tensor lslr_fwd(torch::Tensor arena, torch::Tensor grad, torch::Tensor lr_vec);
 std::vector<torch::Tensor> lslr_bwd(torch::Tensor gout, torch::Tensor grad,
@@ -34,6 +42,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC maxpool 2x2 fwd");
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC maxpool 2x2 bwd");
   m.def("ce_fwd", &ce_fwd, "fused softmax-CE fwd");
+  m.def("ce_dbwd", &ce_dbwd, "softmax-CE double-backward");
+  m.def("bn_act_dbwd", &bn_act_dbwd, "BN+leakyReLU analytic double-backward");
   m.def("ce_bwd", &ce_bwd, "fused softmax-CE bwd");
   m.def("lslr_fwd", &lslr_fwd, "fused LSLR arena update fwd");
   m.def("lslr_bwd", &lslr_bwd, "fused LSLR arena update bwd");

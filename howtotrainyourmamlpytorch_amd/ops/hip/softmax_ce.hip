@@ -53,6 +53,38 @@ __global__ void ce_bwd_kernel(const float* __restrict__ probs,
   }
 }
 
+// Double-backward: dlogits = (p - onehot) * gt/M with p = softmax(logits).
+// Given gdl = dL/d(dlogits):
+//   d_logits = gt/M * (p .* gdl - p * <p, gdl>_row)     (softmax Jacobian)
+//   d_gtask[t] = sum_rows <gdl, p - onehot> / M
+__global__ void ce_dbwd_kernel(const float* __restrict__ probs,
+                               const long* __restrict__ labels,
+                               const float* __restrict__ gdl,
+                               const float* __restrict__ gtask,
+                               float* __restrict__ d_logits,
+                               float* __restrict__ d_gtask,  // pre-zeroed [T]
+                               int T, int M, int ways) {
+  const long rows = (long)T * M;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long wave_id = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const long nwaves = grid_stride() / WAVE;
+  for (long row = wave_id; row < rows; row += nwaves) {
+    const long t = row / M;
+    const float gt_over_M = gtask[t] / (float)M;
+    const float p = (lane < ways) ? probs[row * ways + lane] : 0.f;
+    const float g = (lane < ways) ? gdl[row * ways + lane] : 0.f;
+    const float dot = wave_reduce_sum(p * g);
+    const float dotb = __shfl(dot, 0, WAVE);
+    if (lane < ways) {
+      d_logits[row * ways + lane] = gt_over_M * (p * g - p * dotb);
+    }
+    if (lane == 0) {
+      const float gl = gdl[row * ways + labels[row]];
+      atomicAdd(&d_gtask[t], (dotb - gl) / (float)M);
+    }
+  }
+}
+
 static int grid_for(long total, int threads) {
   long blocks = (total + threads - 1) / threads;
   return (int)std::min<long>(blocks, 4096);
@@ -74,6 +106,25 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels) {
                      lab.data_ptr<long>(), probs.data_ptr<float>(),
                      loss.data_ptr<float>(), T, M, ways);
   return {loss, probs};
+}
+
+std::vector<torch::Tensor> ce_dbwd(torch::Tensor probs, torch::Tensor labels,
+                                   torch::Tensor gdl, torch::Tensor gtask) {
+  const int T = (int)probs.size(0), M = (int)probs.size(1),
+            ways = (int)probs.size(2);
+  auto d_logits = torch::empty_like(probs);
+  auto d_gtask = torch::zeros({T}, probs.options());
+  auto lab = labels.contiguous().to(torch::kLong);
+  auto gdlc = gdl.contiguous().to(torch::kFloat32);
+  auto gtc = gtask.contiguous().to(torch::kFloat32);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const long rows = (long)T * M;
+  hipLaunchKernelGGL(ce_dbwd_kernel, dim3(grid_for(rows * WAVE, 256)),
+                     dim3(256), 0, stream.stream(), probs.data_ptr<float>(),
+                     lab.data_ptr<long>(), gdlc.data_ptr<float>(),
+                     gtc.data_ptr<float>(), d_logits.data_ptr<float>(),
+                     d_gtask.data_ptr<float>(), T, M, ways);
+  return {d_logits, d_gtask};
 }
 
 torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor labels,

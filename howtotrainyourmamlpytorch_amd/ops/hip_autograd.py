@@ -34,9 +34,41 @@ def _ext():
 # ---------------------------------------------------------------------------
 # fused BN + leaky-ReLU
 # ---------------------------------------------------------------------------
+class _BNBwdFn(torch.autograd.Function):
+    """The first backward of BN+act as its own Function: forward = fused
+    bwd kernels; backward = the analytic double-backward kernels
+    (bn_dbwd.hip).  This is what makes second-order MAML's create_graph
+    pass run entirely on custom kernels."""
+
+    @staticmethod
+    def forward(ctx, x3, gamma, beta, u, mean, rstd, slope, act):
+        dx, dgamma_t, dbeta_t = _ext().bn_act_bwd(
+            u, x3, mean, rstd, gamma.float(), beta.float(), slope, act)
+        ctx.save_for_backward(x3, gamma, beta, u, mean, rstd)
+        ctx.slope, ctx.act = slope, act
+        return dx, dgamma_t, dbeta_t
+
+    @staticmethod
+    def backward(ctx, gx, ggam_t, gbet_t):
+        x3, gamma, beta, u, mean, rstd = ctx.saved_tensors
+        T, M, C = x3.shape
+        if ggam_t is None:
+            ggam_t = torch.zeros(T, C, device=x3.device, dtype=torch.float32)
+        if gbet_t is None:
+            gbet_t = torch.zeros(T, C, device=x3.device, dtype=torch.float32)
+        d_u, d_x, sums = _ext().bn_act_dbwd(
+            x3, u, gx.contiguous(), mean, rstd, gamma.float(), beta.float(),
+            ggam_t, gbet_t, ctx.slope, ctx.act)
+        # d_gamma_t = rstd * M * (Gu - s1*G - s2*Gxh)   (means over M)
+        s = sums / M   # [T,5,C] means: {0:s1, 1:s2, 2:G, 3:Gxh, 4:Gu}
+        d_gamma_t = rstd * M * (s[:, 4] - s[:, 0] * s[:, 2] - s[:, 1] * s[:, 3])
+        per_task = gamma.dim() == 2
+        d_gamma = d_gamma_t if per_task else d_gamma_t.sum(0)
+        return (d_x, d_gamma.to(gamma.dtype), None, d_u, None, None, None, None)
+
+
 class _BNActFn(torch.autograd.Function):
-    """x3 [T, M, C] -> (y, mean, var).  Fused kernels forward and on plain
-    backward; differentiable torch composition under create_graph."""
+    """x3 [T, M, C] -> (y, mean, var).  Fused kernels at every order."""
 
     @staticmethod
     def forward(ctx, x3, gamma, beta, eps, slope, act):
@@ -44,36 +76,15 @@ class _BNActFn(torch.autograd.Function):
                                                eps, slope, act)
         ctx.save_for_backward(x3, mean, rstd, gamma, beta)
         ctx.eps, ctx.slope, ctx.act = eps, slope, act
+        ctx.mark_non_differentiable(mean, var)
         return y, mean, var
 
     @staticmethod
     def backward(ctx, dy, dmean, dvar):
         x3, mean, rstd, gamma, beta = ctx.saved_tensors
-        per_task = gamma.dim() == 2
-        if torch.is_grad_enabled():
-            T, M, C = x3.shape
-            xf = x3.float()
-            mu = xf.mean(dim=1, keepdim=True)
-            var_t = xf.var(dim=1, unbiased=False, keepdim=True)
-            inv = torch.rsqrt(var_t + ctx.eps)
-            xhat = (xf - mu) * inv
-            g = gamma.float().view(T if per_task else 1, 1, C)
-            b = beta.float().view(T if per_task else 1, 1, C)
-            dy_eff = dy.float()
-            if ctx.act:
-                pre = xhat * g + b
-                dy_eff = dy_eff * torch.where(
-                    pre > 0, torch.ones_like(pre), torch.full_like(pre, ctx.slope))
-            s1 = dy_eff.mean(dim=1, keepdim=True)
-            s2 = (dy_eff * xhat).mean(dim=1, keepdim=True)
-            dx = (g * inv * (dy_eff - s1 - xhat * s2)).to(x3.dtype)
-            dgamma_t = (dy_eff * xhat).sum(dim=1)
-            dbeta_t = dy_eff.sum(dim=1)
-        else:
-            dx, dgamma_t, dbeta_t = _ext().bn_act_bwd(
-                dy, x3, mean, rstd, gamma.float(), beta.float(),
-                ctx.slope, ctx.act)
-        if per_task:
+        dx, dgamma_t, dbeta_t = _BNBwdFn.apply(
+            x3, gamma, beta, dy.contiguous(), mean, rstd, ctx.slope, ctx.act)
+        if gamma.dim() == 2:
             dgamma, dbeta = dgamma_t, dbeta_t
         else:
             dgamma, dbeta = dgamma_t.sum(0), dbeta_t.sum(0)
@@ -140,6 +151,24 @@ def task_maxpool2x2(x):
 # ---------------------------------------------------------------------------
 # fused softmax-CE
 # ---------------------------------------------------------------------------
+class _CEBwdFn(torch.autograd.Function):
+    """First backward of softmax-CE as a Function: forward = fused bwd
+    kernel; backward = softmax-Jacobian double-backward kernel."""
+
+    @staticmethod
+    def forward(ctx, logits, gtask, probs, labels):
+        ctx.save_for_backward(probs, labels, gtask)
+        ctx.out_dtype = logits.dtype
+        return _ext().ce_bwd(probs, labels, gtask.contiguous()).to(logits.dtype)
+
+    @staticmethod
+    def backward(ctx, gdl):
+        probs, labels, gtask = ctx.saved_tensors
+        d_logits, d_gtask = _ext().ce_dbwd(probs, labels, gdl.contiguous(),
+                                           gtask.contiguous())
+        return d_logits.to(ctx.out_dtype), d_gtask, None, None
+
+
 class _CEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels):
@@ -150,14 +179,7 @@ class _CEFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, gtask):
         logits, probs, labels = ctx.saved_tensors
-        if torch.is_grad_enabled():
-            T, M, ways = logits.shape
-            lf = logits.float()
-            p = torch.softmax(lf, dim=-1)
-            onehot = torch.nn.functional.one_hot(labels, ways).float()
-            dlogits = (p - onehot) * (gtask.view(T, 1, 1) / M)
-            return dlogits.to(logits.dtype), None
-        return _ext().ce_bwd(probs, labels, gtask.contiguous()).to(logits.dtype), None
+        return _CEBwdFn.apply(logits, gtask, probs, labels), None
 
 
 def softmax_cross_entropy(logits, labels):
