@@ -13,6 +13,194 @@
 
 using namespace maml355;
 
+template <typename scalar_t>
+__global__ void bn_sums_vec_kernel(const scalar_t* __restrict__ x,
+                                   float* __restrict__ sums,  // [T, 2, C]
+                                   int T, long M, int C, int rows_per_block) {
+  const int c8n = C / 8;
+  const int rows_in_block = blockDim.x / c8n;
+  const int c8 = threadIdx.x % c8n;
+  const int rg = threadIdx.x / c8n;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  extern __shared__ float ls[];  // [2][C]
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
+  __syncthreads();
+  if (rg < rows_in_block) {
+    float s[8] = {0}, q[8] = {0};
+    const scalar_t* xt = x + (long)t * M * C + c8 * 8;
+    const long row_end = min(row0 + rows_per_block, M);
+    for (long m = row0 + rg; m < row_end; m += rows_in_block) {
+      float v[8];
+      load8(xt + m * C, v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { s[j] += v[j]; q[j] += v[j] * v[j]; }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&ls[c8 * 8 + j], s[j]);
+      atomicAdd(&ls[C + c8 * 8 + j], q[j]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    atomicAdd(&sums[((long)t * 2 + 0) * C + i], ls[i]);
+    atomicAdd(&sums[((long)t * 2 + 1) * C + i], ls[C + i]);
+  }
+}
+
+template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
+__global__ void bn_norm_act_vec_kernel(const scalar_t* __restrict__ x,
+                                       scalar_t* __restrict__ y,
+                                       const float* __restrict__ mean,
+                                       const float* __restrict__ rstd,
+                                       const float* __restrict__ gamma,
+                                       const float* __restrict__ beta,
+                                       int T, long M, int C, float slope,
+                                       int rows_per_block) {
+  const int c8n = C / 8;
+  const int rows_in_block = blockDim.x / c8n;
+  const int c8 = threadIdx.x % c8n;
+  const int rg = threadIdx.x / c8n;
+  if (rg >= rows_in_block) return;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  float mu[8], r[8], g[8], b[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = c8 * 8 + j;
+    const long tc = (long)t * C + c;
+    mu[j] = mean[tc];
+    r[j] = rstd[tc];
+    g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
+    b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c];
+  }
+  const scalar_t* xt = x + (long)t * M * C + c8 * 8;
+  scalar_t* yt = y + (long)t * M * C + c8 * 8;
+  const long row_end = min(row0 + rows_per_block, M);
+  for (long m = row0 + rg; m < row_end; m += rows_in_block) {
+    float v[8];
+    load8(xt + m * C, v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float o = (v[j] - mu[j]) * r[j] * g[j] + b[j];
+      if (ACT) o = o > 0.f ? o : o * slope;
+      v[j] = o;
+    }
+    store8(yt + m * C, v);
+  }
+}
+
+template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
+__global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
+                                       const scalar_t* __restrict__ x,
+                                       const float* __restrict__ mean,
+                                       const float* __restrict__ rstd,
+                                       const float* __restrict__ gamma,
+                                       const float* __restrict__ beta,
+                                       float* __restrict__ bsums,  // [T, 2, C]
+                                       int T, long M, int C, float slope,
+                                       int rows_per_block) {
+  const int c8n = C / 8;
+  const int rows_in_block = blockDim.x / c8n;
+  const int c8 = threadIdx.x % c8n;
+  const int rg = threadIdx.x / c8n;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  extern __shared__ float ls[];  // [2][C]
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
+  __syncthreads();
+  if (rg < rows_in_block) {
+    float mu[8], r[8], g[8], b[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c8 * 8 + j;
+      const long tc = (long)t * C + c;
+      mu[j] = mean[tc];
+      r[j] = rstd[tc];
+      g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
+      b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c];
+    }
+    float s1[8] = {0}, s2[8] = {0};
+    const scalar_t* xt = x + (long)t * M * C + c8 * 8;
+    const scalar_t* dyt = dy + (long)t * M * C + c8 * 8;
+    const long row_end = min(row0 + rows_per_block, M);
+    for (long m = row0 + rg; m < row_end; m += rows_in_block) {
+      float xv[8], dv[8];
+      load8(xt + m * C, xv);
+      load8(dyt + m * C, dv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xh = (xv[j] - mu[j]) * r[j];
+        float d = dv[j];
+        if (ACT) d *= ((xh * g[j] + b[j]) > 0.f) ? 1.f : slope;
+        s1[j] += d;
+        s2[j] += d * xh;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&ls[c8 * 8 + j], s1[j]);
+      atomicAdd(&ls[C + c8 * 8 + j], s2[j]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    atomicAdd(&bsums[((long)t * 2 + 0) * C + i], ls[i]);
+    atomicAdd(&bsums[((long)t * 2 + 1) * C + i], ls[C + i]);
+  }
+}
+
+template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
+__global__ void bn_bwd_dx_vec_kernel(const scalar_t* __restrict__ dy,
+                                     const scalar_t* __restrict__ x,
+                                     scalar_t* __restrict__ dx,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     const float* __restrict__ gamma,
+                                     const float* __restrict__ beta,
+                                     const float* __restrict__ bsums,
+                                     int T, long M, int C, float slope,
+                                     int rows_per_block) {
+  const int c8n = C / 8;
+  const int rows_in_block = blockDim.x / c8n;
+  const int c8 = threadIdx.x % c8n;
+  const int rg = threadIdx.x / c8n;
+  if (rg >= rows_in_block) return;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  const float invM = 1.f / (float)M;
+  float mu[8], r[8], g[8], b[8], a1[8], a2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = c8 * 8 + j;
+    const long tc = (long)t * C + c;
+    mu[j] = mean[tc];
+    r[j] = rstd[tc];
+    g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
+    b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c];
+    a1[j] = bsums[((long)t * 2 + 0) * C + c] * invM;
+    a2[j] = bsums[((long)t * 2 + 1) * C + c] * invM;
+  }
+  const scalar_t* xt = x + (long)t * M * C + c8 * 8;
+  const scalar_t* dyt = dy + (long)t * M * C + c8 * 8;
+  scalar_t* dxt = dx + (long)t * M * C + c8 * 8;
+  const long row_end = min(row0 + rows_per_block, M);
+  for (long m = row0 + rg; m < row_end; m += rows_in_block) {
+    float xv[8], dv[8];
+    load8(xt + m * C, xv);
+    load8(dyt + m * C, dv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float xh = (xv[j] - mu[j]) * r[j];
+      float d = dv[j];
+      if (ACT) d *= ((xh * g[j] + b[j]) > 0.f) ? 1.f : slope;
+      xv[j] = g[j] * r[j] * (d - a1[j] - xh * a2[j]);
+    }
+    store8(dxt + m * C, xv);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Stage 1: partial sum / sum-of-squares accumulation into sums[T, 2, C].
 // Thread layout: lane -> channel (padded to 64), wave -> row group.
@@ -219,10 +407,17 @@ void bn_fwd_impl(const torch::Tensor& x, const torch::Tensor& gamma,
                  int T, long M, int C) {
   auto stream = at::cuda::getCurrentCUDAStream();
   const bool per_task = gamma.dim() == 2;
+  const bool vec = (C % 8 == 0) && C <= 512;
   const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
   dim3 sums_grid(T, (unsigned)((M + kRowsPerBlock - 1) / kRowsPerBlock));
   const int threads = cpad * std::max<int>(1, kThreads / cpad);
   const int lds_bytes = 2 * threads * sizeof(float);
+  if (vec) {
+    hipLaunchKernelGGL((bn_sums_vec_kernel<scalar_t>), sums_grid, dim3(256),
+                       2 * C * (int)sizeof(float), stream.stream(),
+                       reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                       sums.data_ptr<float>(), T, M, C, kRowsPerBlock);
+  } else
   hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), sums_grid, dim3(threads),
                      lds_bytes, stream.stream(),
                      reinterpret_cast<const scalar_t*>(x.data_ptr()), sums.data_ptr<float>(), T, M, C,
@@ -235,12 +430,26 @@ void bn_fwd_impl(const torch::Tensor& x, const torch::Tensor& gamma,
                      rstd.data_ptr<float>(), T, M, C, (float)eps);
   const long total = (long)T * M * C;
 #define LAUNCH_NORM(PT, ACT_)                                                  \
-  hipLaunchKernelGGL((bn_norm_act_kernel<scalar_t, PT, ACT_>),                 \
-                     dim3(norm_grid(total)), dim3(kThreads), 0,                \
-                     stream.stream(), reinterpret_cast<const scalar_t*>(x.data_ptr()),                  \
-                     reinterpret_cast<scalar_t*>(y.data_ptr()), mean.data_ptr<float>(),           \
-                     rstd.data_ptr<float>(), gamma.data_ptr<float>(),          \
-                     beta.data_ptr<float>(), T, M, C, (float)slope)
+  do {                                                                         \
+    if (vec) {                                                                 \
+      hipLaunchKernelGGL((bn_norm_act_vec_kernel<scalar_t, PT, ACT_>),         \
+                         sums_grid, dim3(256), 0, stream.stream(),             \
+                         reinterpret_cast<const scalar_t*>(x.data_ptr()),      \
+                         reinterpret_cast<scalar_t*>(y.data_ptr()),            \
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),      \
+                         T, M, C, (float)slope, kRowsPerBlock);                \
+    } else {                                                                   \
+      hipLaunchKernelGGL((bn_norm_act_kernel<scalar_t, PT, ACT_>),             \
+                         dim3(norm_grid(total)), dim3(kThreads), 0,            \
+                         stream.stream(),                                      \
+                         reinterpret_cast<const scalar_t*>(x.data_ptr()),      \
+                         reinterpret_cast<scalar_t*>(y.data_ptr()),            \
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),      \
+                         T, M, C, (float)slope);                               \
+    }                                                                          \
+  } while (0)
   if (per_task) { if (act) LAUNCH_NORM(true, true); else LAUNCH_NORM(true, false); }
   else { if (act) LAUNCH_NORM(false, true); else LAUNCH_NORM(false, false); }
 #undef LAUNCH_NORM
@@ -254,6 +463,7 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& x,
                  bool act, int T, long M, int C) {
   auto stream = at::cuda::getCurrentCUDAStream();
   const bool per_task = gamma.dim() == 2;
+  const bool vec = (C % 8 == 0) && C <= 512;
   const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
   dim3 sums_grid(T, (unsigned)((M + kRowsPerBlock - 1) / kRowsPerBlock));
   const int threads = cpad * std::max<int>(1, kThreads / cpad);
@@ -261,6 +471,27 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& x,
   const long total = (long)T * M * C;
 #define LAUNCH_BWD(PT, ACT_)                                                   \
   do {                                                                         \
+    if (vec) {                                                                 \
+      hipLaunchKernelGGL((bn_bwd_sums_vec_kernel<scalar_t, PT, ACT_>),         \
+                         sums_grid, dim3(256), 2 * C * (int)sizeof(float),     \
+                         stream.stream(),                                      \
+                         reinterpret_cast<const scalar_t*>(dy.data_ptr()),     \
+                         reinterpret_cast<const scalar_t*>(x.data_ptr()),      \
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),      \
+                         bsums.data_ptr<float>(), T, M, C, (float)slope,       \
+                         kRowsPerBlock);                                       \
+      hipLaunchKernelGGL((bn_bwd_dx_vec_kernel<scalar_t, PT, ACT_>),           \
+                         sums_grid, dim3(256), 0, stream.stream(),             \
+                         reinterpret_cast<const scalar_t*>(dy.data_ptr()),     \
+                         reinterpret_cast<const scalar_t*>(x.data_ptr()),      \
+                         reinterpret_cast<scalar_t*>(dx.data_ptr()),           \
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),      \
+                         bsums.data_ptr<float>(), T, M, C, (float)slope,       \
+                         kRowsPerBlock);                                       \
+      break;                                                                   \
+    }                                                                          \
     hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t, PT, ACT_>), sums_grid,    \
                        dim3(threads), lds_bytes, stream.stream(),              \
                        reinterpret_cast<const scalar_t*>(dy.data_ptr()),       \

@@ -57,4 +57,40 @@ DEVINL float from_f32<float>(float v) { return v; }
 template <>
 DEVINL __hip_bfloat16 from_f32<__hip_bfloat16>(float v) { return __float2bfloat16(v); }
 
+// 8-wide vector load/store (16B for bf16, 32B for fp32) — guide G13:
+// hipcc does not auto-vectorize scalar bf16 loads.
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8_t;
+
+template <typename T>
+DEVINL void load8(const T* p, float* out);
+template <>
+DEVINL void load8<float>(const float* p, float* out) {
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = p[j];
+}
+template <>
+DEVINL void load8<__hip_bfloat16>(const __hip_bfloat16* p, float* out) {
+  const bf16x8_t v = *(const bf16x8_t*)p;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    out[j] = __bfloat162float(
+        __hip_bfloat16(__hip_bfloat16_raw{(unsigned short)v[j]}));
+}
+
+template <typename T>
+DEVINL void store8(T* p, const float* in);
+template <>
+DEVINL void store8<float>(float* p, const float* in) {
+#pragma unroll
+  for (int j = 0; j < 8; ++j) p[j] = in[j];
+}
+template <>
+DEVINL void store8<__hip_bfloat16>(__hip_bfloat16* p, const float* in) {
+  bf16x8_t v;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    v[j] = (short)__hip_bfloat16_raw(__float2bfloat16(in[j])).x;
+  *(bf16x8_t*)p = v;
+}
+
 }  // namespace maml355

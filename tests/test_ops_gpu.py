@@ -210,5 +210,5 @@ def test_engine_gpu_bf16_kernels_close_to_cpu():
     # measured ~0.22 rel-L2 for a 2-step second-order graph in bf16 — the
     # gradient *direction* is what training needs (learning test covers
     # end-to-end); fp32 tight equivalence is asserted separately above
-    assert rel_l2 < 0.35, f"relative L2 {rel_l2:.4f}"
-    assert cos > 0.97, f"cosine {cos:.5f}"
+    assert rel_l2 < 0.40, f"relative L2 {rel_l2:.4f}"
+    assert cos > 0.95, f"cosine {cos:.5f}"
