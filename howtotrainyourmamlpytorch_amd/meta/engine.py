@@ -248,19 +248,93 @@ class MAMLFewShotClassifier(nn.Module):
         return losses, per_task_preds
 
     # ------------------------------------------------------------------
-    # checkpointing — same file layout as the reference: one torch pickle
-    # per epoch holding the whole experiment ``state`` dict with
-    # state['network'] = self.state_dict() (``few_shot_learning_system.py:399-424``)
+    # checkpointing — same file layout AND key naming as the reference: one
+    # torch pickle per epoch holding the experiment ``state`` dict with
+    # state['network'] using reference-style names
+    # (``few_shot_learning_system.py:399-424``):
+    #   classifier.layer_dict.conv{i}.conv.{weight,bias}
+    #   classifier.layer_dict.conv{i}.norm_layer.{weight,bias,running_*}
+    #   classifier.layer_dict.linear.{weights,bias}
+    #   inner_loop_optimizer.names_learning_rates_dict.layer_dict-...-weight
+    # The flat arena is expanded on save and re-packed on load; the linear
+    # weight is permuted between our NHWC flatten order and the reference's
+    # NCHW flatten order so checkpoints are interchangeable.
+    def _linear_to_reference(self, w: torch.Tensor) -> torch.Tensor:
+        cls = self.classifier
+        if not cls.max_pooling:
+            return w
+        h, w_sp = cls.final_spatial
+        ways = w.shape[0]
+        return (w.view(ways, h, w_sp, cls.num_filters).permute(0, 3, 1, 2)
+                .reshape(ways, -1))
+
+    def _linear_from_reference(self, w: torch.Tensor) -> torch.Tensor:
+        cls = self.classifier
+        if not cls.max_pooling:
+            return w
+        h, w_sp = cls.final_spatial
+        ways = w.shape[0]
+        return (w.view(ways, cls.num_filters, h, w_sp).permute(0, 2, 3, 1)
+                .reshape(ways, -1))
+
+    def reference_state_dict(self) -> Dict[str, torch.Tensor]:
+        cls = self.classifier
+        out: Dict[str, torch.Tensor] = {}
+        views = cls.arena.views(self.classifier.theta.detach())
+        for name, v in views.items():
+            t = v.clone()
+            if name == "layer_dict.linear.weights":
+                t = self._linear_to_reference(t)
+            out["classifier." + name] = t
+        if cls.norm_layer_type == "batch_norm":
+            for i in range(cls.num_stages):
+                prefix = f"classifier.layer_dict.conv{i}.norm_layer."
+                if not cls.inner_loop_bn_params:
+                    out[prefix + "weight"] = getattr(cls, f"bn_weight_{i}").detach().clone()
+                    out[prefix + "bias"] = getattr(cls, f"bn_bias_{i}").detach().clone()
+                out[prefix + "running_mean"] = getattr(cls, f"bn_running_mean_{i}").clone()
+                out[prefix + "running_var"] = getattr(cls, f"bn_running_var_{i}").clone()
+        for spec in cls.arena.specs:
+            key = ("inner_loop_optimizer.names_learning_rates_dict."
+                   + spec.name.replace(".", "-"))
+            out[key] = self.inner_loop_lrs.detach()[spec.slot].clone()
+        return out
+
+    def load_reference_state_dict(self, sd: Dict[str, torch.Tensor]) -> None:
+        cls = self.classifier
+        named = {}
+        for spec in cls.arena.specs:
+            t = sd["classifier." + spec.name]
+            if spec.name == "layer_dict.linear.weights":
+                t = self._linear_from_reference(t)
+            named[spec.name] = t
+        with torch.no_grad():
+            self.classifier.theta.copy_(cls.arena.pack(named).to(self.device))
+            if cls.norm_layer_type == "batch_norm":
+                for i in range(cls.num_stages):
+                    prefix = f"classifier.layer_dict.conv{i}.norm_layer."
+                    if not cls.inner_loop_bn_params and prefix + "weight" in sd:
+                        getattr(cls, f"bn_weight_{i}").copy_(sd[prefix + "weight"])
+                        getattr(cls, f"bn_bias_{i}").copy_(sd[prefix + "bias"])
+                    if prefix + "running_mean" in sd:
+                        getattr(cls, f"bn_running_mean_{i}").copy_(sd[prefix + "running_mean"])
+                        getattr(cls, f"bn_running_var_{i}").copy_(sd[prefix + "running_var"])
+            for spec in cls.arena.specs:
+                key = ("inner_loop_optimizer.names_learning_rates_dict."
+                       + spec.name.replace(".", "-"))
+                if key in sd:
+                    self.inner_loop_lrs[spec.slot].copy_(sd[key])
+
     def save_model(self, model_save_dir: str, state: Dict) -> None:
         state = dict(state)
-        state["network"] = self.state_dict()
+        state["network"] = self.reference_state_dict()
         state["optimizer"] = self.optimizer.state_dict()
         torch.save(state, f=model_save_dir)
 
     def load_model(self, model_save_dir: str, model_name: str, model_idx) -> Dict:
         filepath = os.path.join(model_save_dir, f"{model_name}_{model_idx}")
         state = torch.load(filepath, map_location=self.device, weights_only=False)
-        self.load_state_dict(state_dict=state["network"])
+        self.load_reference_state_dict(state["network"])
         if "optimizer" in state:
             self.optimizer.load_state_dict(state["optimizer"])
         return state
