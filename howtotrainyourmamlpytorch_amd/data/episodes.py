@@ -85,6 +85,8 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
         self.current_set = current_set
         self.dataset_name = args.dataset_name
 
+        self._npz_images = None  # [num_classes, spc, H, W] uint8 when npz-backed
+        self._npz_class_index: Dict[str, int] = {}
         self.datasets = self._load_datapaths()
         # seeds per set: val/test share the fixed val seed (reference uses
         # the same seed for val and test streams, data.py:141-142)
@@ -93,7 +95,7 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
         self.seed = dict(self.init_seed)
 
         self._memory: Optional[Dict[str, List[np.ndarray]]] = None
-        if getattr(args, "load_into_memory", False):
+        if getattr(args, "load_into_memory", False) and self._npz_images is None:
             self._preload()
 
     # ------------------------------------------------------------------
@@ -112,6 +114,15 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
             except (json.JSONDecodeError, OSError):
                 pass
         root = self.args.dataset_path
+        npz_path = None
+        if str(root).endswith(".npz"):
+            npz_path = root
+        elif getattr(self.args, "load_from_npz_files", False):
+            cand = str(root).rstrip(os.sep) + ".npz"
+            if os.path.isfile(cand):
+                npz_path = cand
+        if npz_path is not None:
+            return self._load_npz(npz_path)
         if not os.path.isdir(root):
             raise FileNotFoundError(
                 f"dataset_path {root!r} does not exist (set DATASET_DIR or "
@@ -132,6 +143,28 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
         return sets
 
     # ------------------------------------------------------------------
+    def _load_npz(self, npz_path: str) -> Dict[str, Dict[str, List[int]]]:
+        """Compact preprocessed dataset (tools/make_omniglot_npz.py):
+        images [num_classes, samples_per_class, H, W] uint8 + class names.
+        Per-class 'file lists' become sample-index lists."""
+        data = np.load(npz_path, allow_pickle=True)
+        self._npz_images = data["images"]
+        names = [str(n) for n in data["class_names"]]
+        self._npz_class_index = {n: i for i, n in enumerate(names)}
+        spc = self._npz_images.shape[1]
+        classes = {n: list(range(spc)) for n in names}
+        return split_classes(classes, self.args.train_val_test_split)
+
+    def _npz_image(self, cname: str, index: int) -> np.ndarray:
+        arr = self._npz_images[self._npz_class_index[cname], index]
+        if arr.shape != (self.image_height, self.image_width):
+            from PIL import Image
+            img = Image.fromarray(arr, mode="L").resize(
+                (self.image_width, self.image_height), Image.LANCZOS)
+            arr = np.asarray(img)
+        out = arr.astype(np.float32) / 255.0
+        return out[:, :, None]
+
     def _load_image(self, path: str) -> np.ndarray:
         """Decode + resize one image -> float32 HWC in [0, 1]
         (reference: ``data.py:374-395``)."""
@@ -165,6 +198,8 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
         self._memory = store
 
     def _get_image(self, cname: str, index: int) -> np.ndarray:
+        if self._npz_images is not None:
+            return self._npz_image(cname, index)
         if self._memory is not None and cname in self._memory:
             return self._memory[cname][index]
         return self._load_image(self.datasets[self.current_set][cname][index])

@@ -1,0 +1,37 @@
+"""Preprocess an Omniglot class-folder tree into a compact 28x28 uint8 npz
+(images [num_classes, samples_per_class, 28, 28] + class_names), the format
+consumed by FewShotEpisodeDataset's npz path.  Run once at dev time:
+
+    python tools/make_omniglot_npz.py <omniglot_root> datasets/omniglot_28x28.npz
+"""
+import os
+import sys
+
+import numpy as np
+from PIL import Image
+
+
+def main(root, out):
+    classes = {}
+    for dirpath, _dirs, files in os.walk(root):
+        pngs = sorted(f for f in files if f.endswith(".png"))
+        if not pngs:
+            continue
+        parts = os.path.normpath(dirpath).split(os.sep)
+        cname = os.sep.join(parts[-2:])
+        classes[cname] = [os.path.join(dirpath, f) for f in pngs]
+    names = sorted(classes)
+    spc = min(len(classes[c]) for c in names)
+    images = np.zeros((len(names), spc, 28, 28), dtype=np.uint8)
+    for ci, cname in enumerate(names):
+        for si, path in enumerate(classes[cname][:spc]):
+            img = Image.open(path).convert("L").resize((28, 28), Image.LANCZOS)
+            images[ci, si] = np.asarray(img, dtype=np.uint8)
+    np.savez_compressed(out, images=images,
+                        class_names=np.array(names, dtype=object))
+    print(f"{len(names)} classes x {spc} samples -> {out} "
+          f"({os.path.getsize(out)/1e6:.1f} MB)")
+
+
+if __name__ == "__main__":
+    main(sys.argv[1], sys.argv[2])
