@@ -124,7 +124,9 @@ class MAMLFewShotClassifier(nn.Module):
                      y_support [B,N,S], y_target [B,N,T])
         Returns (losses dict, per_task_target_preds [B, N*T, ways]).
         """
-        x_support, x_target, y_support, y_target = data_batch
+        # real episode batches carry the per-episode seed as a 5th element
+        # (reference data.py:478-524); synthetic batches are 4-tuples
+        x_support, x_target, y_support, y_target = data_batch[:4]
         x_support = x_support.to(self.device, non_blocking=True)
         x_target = x_target.to(self.device, non_blocking=True)
         y_support = y_support.to(self.device, non_blocking=True)

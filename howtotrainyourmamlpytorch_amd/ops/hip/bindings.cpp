@@ -33,7 +33,8 @@ torch::Tensor tconv_repack(torch::Tensor w, bool dgrad);
 torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
                        c10::optional<torch::Tensor> bias, long pad,
                        long Ho, long Wo);
-torch::Tensor tconv_wgrad(torch::Tensor dy, torch::Tensor x, long pad);
+std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
+                                       long pad, bool with_bias);
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -256,6 +256,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 __global__ __launch_bounds__(256, 2)
 void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                         float* __restrict__ dWacc,
+                        float* __restrict__ dBacc,  // [T, F] or nullptr
                         int T, int NB, int H, int W, int C,
                         int Ho, int Wo, int F, int pad) {
   const int t = blockIdx.z;
@@ -295,6 +296,12 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   f32x4 acc[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  // fused bias-grad: only the n0==0 column block accumulates (the dY tile
+  // is identical across column blocks)
+  const bool do_bias = (dBacc != nullptr) && (blockIdx.x == 0);
+  float db_acc = 0.f;
+  const int db_f = threadIdx.x & 63;
+  const int db_q = threadIdx.x >> 6;        // quarter of the k range
 
   for (long k0 = kchunk0; k0 < kchunk_end; k0 += WBK) {
     __syncthreads();
@@ -377,6 +384,23 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
           acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
         }
       }
+    }
+    if (do_bias && db_f < F) {
+#pragma unroll
+      for (int kk = db_q * 16; kk < db_q * 16 + 16; ++kk) {
+        db_acc += __bfloat162float(
+            __hip_bfloat16(__hip_bfloat16_raw{(unsigned short)lds_at[db_f][kk]}));
+      }
+    }
+  }
+  if (do_bias) {
+    __shared__ float db_lds[4][64];
+    db_lds[db_q][db_f] = db_acc;
+    __syncthreads();
+    if (db_q == 0 && db_f < F) {
+      atomicAdd(&dBacc[(long)t * F + db_f],
+                db_lds[0][db_f] + db_lds[1][db_f] + db_lds[2][db_f] +
+                    db_lds[3][db_f]);
     }
   }
 
@@ -473,8 +497,11 @@ torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
   return y;
 }
 
-// dy [T, NB, Ho, Wo, F] bf16 ; x [T, NB, H, W, C] bf16 -> dw [T, F, C, 3, 3] fp32
-torch::Tensor tconv_wgrad(torch::Tensor dy, torch::Tensor x, long pad) {
+// dy [T, NB, Ho, Wo, F] bf16 ; x [T, NB, H, W, C] bf16
+// -> {dw [T, F, C, 3, 3] fp32, db [T, F] fp32}   (db fused — the dY tile
+// is already staged for the GEMM, so the bias reduction is nearly free)
+std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
+                                       long pad, bool with_bias) {
   TORCH_CHECK(dy.is_cuda() && x.is_cuda());
   auto dyc = dy.contiguous();
   auto xc = x.contiguous();
@@ -484,6 +511,7 @@ torch::Tensor tconv_wgrad(torch::Tensor dy, torch::Tensor x, long pad) {
   TORCH_CHECK(F <= 64, "F must be <= 64");
   const int N9 = 9 * C;
   auto acc = torch::zeros({T, N9, F}, x.options().dtype(torch::kFloat32));
+  auto db = torch::zeros({T, F}, x.options().dtype(torch::kFloat32));
   const long Ktot = (long)NB * Ho * Wo;
   dim3 grid((unsigned)((N9 + 63) / 64),
             (unsigned)((Ktot + WG_KCHUNK - 1) / WG_KCHUNK), T);
@@ -491,13 +519,15 @@ torch::Tensor tconv_wgrad(torch::Tensor dy, torch::Tensor x, long pad) {
   hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(256), 0, stream.stream(),
                      reinterpret_cast<const bf16*>(dyc.data_ptr()),
                      reinterpret_cast<const bf16*>(xc.data_ptr()),
-                     acc.data_ptr<float>(), T, NB, H, W, C, Ho, Wo, F, (int)pad);
+                     acc.data_ptr<float>(),
+                     with_bias ? db.data_ptr<float>() : nullptr,
+                     T, NB, H, W, C, Ho, Wo, F, (int)pad);
   auto dw = torch::empty({T, F, C, 3, 3}, x.options().dtype(torch::kFloat32));
   const long total = (long)T * F * C * 9;
   hipLaunchKernelGGL(wgrad_finalize_kernel, dim3(ew_grid2(total, 256)),
                      dim3(256), 0, stream.stream(), acc.data_ptr<float>(),
                      dw.data_ptr<float>(), T, F, C);
-  return dw;
+  return {dw, db};
 }
 
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B) {

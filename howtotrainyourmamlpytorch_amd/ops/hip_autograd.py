@@ -235,8 +235,11 @@ class _ConvFwdFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = _ConvDgradFn.apply(dy, w, ctx.pad)
         if ctx.needs_input_grad[1]:
-            dw = _ConvWgradFn.apply(dy, x, ctx.pad)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+            want_bias = ctx.has_bias and ctx.needs_input_grad[2]
+            dw, db = _ConvWgradFn.apply(dy, x, ctx.pad, want_bias)
+            if not want_bias:
+                db = None
+        elif ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(1, 2, 3))
         return dx, dw, db, None
 
@@ -262,27 +265,36 @@ class _ConvDgradFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             d_dy = _ConvFwdFn.apply(g, w, None, ctx.pad)
         if ctx.needs_input_grad[1]:
-            d_w = _ConvWgradFn.apply(dy, g, ctx.pad)
+            d_w = _ConvWgradFn.apply(dy, g, ctx.pad, False)[0]
         return d_dy, d_w, None
 
 
 class _ConvWgradFn(torch.autograd.Function):
-    @staticmethod
-    def forward(ctx, dy, x, pad):
-        ctx.save_for_backward(dy, x)
-        ctx.pad = pad
-        return _ext().tconv_wgrad(dy, x, pad)
+    """(dy, x) -> (dw, db); db is the fused bias gradient (sum of dy over
+    positions) — the wgrad kernel already stages the dY tiles, so the
+    reduction rides along instead of a separate big ATen sum."""
 
     @staticmethod
-    def backward(ctx, gw):
+    def forward(ctx, dy, x, pad, with_bias):
+        ctx.save_for_backward(dy, x)
+        ctx.pad = pad
+        dw, db = _ext().tconv_wgrad(dy, x, pad, with_bias)
+        return dw, db
+
+    @staticmethod
+    def backward(ctx, gw, gdb):
         dy, x = ctx.saved_tensors
-        gw = gw.contiguous()
         d_dy = d_x = None
         if ctx.needs_input_grad[0]:
-            d_dy = _ConvFwdFn.apply(x, gw, None, ctx.pad)
-        if ctx.needs_input_grad[1]:
-            d_x = _ConvDgradFn.apply(dy, gw, ctx.pad)
-        return d_dy, d_x, None
+            if gw is not None:
+                d_dy = _ConvFwdFn.apply(x, gw.contiguous(), None, ctx.pad)
+            if gdb is not None:
+                T, F = gdb.shape
+                add = gdb.view(T, 1, 1, 1, F).to(dy.dtype)
+                d_dy = add.expand_as(dy).contiguous() if d_dy is None else d_dy + add
+        if ctx.needs_input_grad[1] and gw is not None:
+            d_x = _ConvDgradFn.apply(dy, gw.contiguous(), ctx.pad)
+        return d_dy, d_x, None, None
 
 
 def task_conv3x3(x, w, b=None, stride=1, padding=1):
