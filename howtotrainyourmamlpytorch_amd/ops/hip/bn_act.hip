@@ -413,10 +413,12 @@ void bn_fwd_impl(const torch::Tensor& x, const torch::Tensor& gamma,
   const int threads = cpad * std::max<int>(1, kThreads / cpad);
   const int lds_bytes = 2 * threads * sizeof(float);
   if (vec) {
-    hipLaunchKernelGGL((bn_sums_vec_kernel<scalar_t>), sums_grid, dim3(256),
+    const int rpb = 1024;  // amortize the per-block LDS/global atomic rounds
+    dim3 g(T, (unsigned)((M + rpb - 1) / rpb));
+    hipLaunchKernelGGL((bn_sums_vec_kernel<scalar_t>), g, dim3(256),
                        2 * C * (int)sizeof(float), stream.stream(),
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
-                       sums.data_ptr<float>(), T, M, C, kRowsPerBlock);
+                       sums.data_ptr<float>(), T, M, C, rpb);
   } else
   hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), sums_grid, dim3(threads),
                      lds_bytes, stream.stream(),
@@ -472,15 +474,16 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& x,
 #define LAUNCH_BWD(PT, ACT_)                                                   \
   do {                                                                         \
     if (vec) {                                                                 \
+      dim3 gs(T, (unsigned)((M + 1023) / 1024));                               \
       hipLaunchKernelGGL((bn_bwd_sums_vec_kernel<scalar_t, PT, ACT_>),         \
-                         sums_grid, dim3(256), 2 * C * (int)sizeof(float),     \
+                         gs, dim3(256), 2 * C * (int)sizeof(float),            \
                          stream.stream(),                                      \
                          reinterpret_cast<const scalar_t*>(dy.data_ptr()),     \
                          reinterpret_cast<const scalar_t*>(x.data_ptr()),      \
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
                          gamma.data_ptr<float>(), beta.data_ptr<float>(),      \
                          bsums.data_ptr<float>(), T, M, C, (float)slope,       \
-                         kRowsPerBlock);                                       \
+                         1024);                                                \
       hipLaunchKernelGGL((bn_bwd_dx_vec_kernel<scalar_t, PT, ACT_>),           \
                          sums_grid, dim3(256), 0, stream.stream(),             \
                          reinterpret_cast<const scalar_t*>(dy.data_ptr()),     \

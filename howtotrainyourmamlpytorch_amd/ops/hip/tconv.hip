@@ -106,7 +106,10 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
   __shared__ short lds_a[BM][BK + APAD];
   __shared__ short lds_bt[64][BK + APAD];   // B^T: [col][k]
   __shared__ int row_h[BM], row_w[BM], row_n[BM];
-  __shared__ int ktab_dy[BK], ktab_dx[BK], ktab_c[BK];
+  // full k-table hoisted out of the K-loop (9*Ci <= 576 always, since
+  // Ci <= 64): one decode per k for the whole block, one fewer barrier
+  // per K-step
+  __shared__ int ktab_dy[576 + BK], ktab_dx[576 + BK], ktab_c[576 + BK];
 
   // per-block row decode (once)
   for (int m = threadIdx.x; m < BM; m += blockDim.x) {
@@ -118,6 +121,16 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
       row_h[m] = ho; row_w[m] = wo; row_n[m] = n;
     } else {
       row_n[m] = -1;
+    }
+  }
+  for (int k = threadIdx.x; k < ksteps * BK; k += blockDim.x) {
+    if (k < K9) {
+      const int kyx = k / Ci;
+      ktab_c[k] = k - kyx * Ci;
+      ktab_dy[k] = kyx / 3;
+      ktab_dx[k] = kyx % 3;
+    } else {
+      ktab_c[k] = -1;
     }
   }
 
@@ -133,36 +146,22 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 #pragma unroll
   for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 
+  __syncthreads();
   for (int ks = 0; ks < ksteps; ++ks) {
     const int k0 = ks * BK;
-    // k-table for this step
-    __syncthreads();
-    if (threadIdx.x < BK) {
-      const int k = k0 + (int)threadIdx.x;
-      if (k < K9) {
-        const int kyx = k / Ci;
-        ktab_c[threadIdx.x] = k - kyx * Ci;
-        ktab_dy[threadIdx.x] = kyx / 3;
-        ktab_dx[threadIdx.x] = kyx % 3;
-      } else {
-        ktab_c[threadIdx.x] = -1;
-      }
-    }
-    __syncthreads();
-
     // stage A (im2col): slot s -> (m = s>>3, kk0 = (s&7)*8); vector path
     // when the 8-k run stays inside one (ky,kx) slice (contiguous c).
     for (int s = threadIdx.x; s < BM * 8; s += blockDim.x) {
       const int m = s >> 3;
       const int kk0 = (s & 7) * 8;
       const int n = row_n[m];
-      const int c0 = ktab_c[kk0];
-      const int c7 = ktab_c[kk0 + 7];
+      const int c0 = ktab_c[k0 + kk0];
+      const int c7 = ktab_c[k0 + kk0 + 7];
       if ((Ci % 8 == 0) && c0 >= 0 && c7 == c0 + 7) {
         bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (n >= 0) {
-          const int h = row_h[m] + ktab_dy[kk0] - pad;
-          const int w = row_w[m] + ktab_dx[kk0] - pad;
+          const int h = row_h[m] + ktab_dy[k0 + kk0] - pad;
+          const int w = row_w[m] + ktab_dx[k0 + kk0] - pad;
           if (h >= 0 && h < H && w >= 0 && w < W) {
             v = *(const bf16x8*)&(
                 (const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c0];
@@ -172,11 +171,11 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
       } else {
         for (int j = 0; j < 8; ++j) {
           const int kk = kk0 + j;
-          const int c = ktab_c[kk];
+          const int c = ktab_c[k0 + kk];
           short v = 0;
           if (n >= 0 && c >= 0) {
-            const int h = row_h[m] + ktab_dy[kk] - pad;
-            const int w = row_w[m] + ktab_dx[kk] - pad;
+            const int h = row_h[m] + ktab_dy[k0 + kk] - pad;
+            const int w = row_w[m] + ktab_dx[k0 + kk] - pad;
             if (h >= 0 && h < H && w >= 0 && w < W) {
               v = ((const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c];
             }
@@ -212,16 +211,17 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 
     // fragments + MFMA (2 K-slices per staged tile)
 #pragma unroll
-    for (int ks = 0; ks < BK / 32; ++ks) {
-      bf16x8 a = *(const bf16x8*)&lds_a[wave * 16 + fr][ks * 32 + fk * 8];
+    for (int ksl = 0; ksl < BK / 32; ++ksl) {
+      bf16x8 a = *(const bf16x8*)&lds_a[wave * 16 + fr][ksl * 32 + fk * 8];
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         if (nt < ntiles) {
-          bf16x8 b = *(const bf16x8*)&lds_bt[nt * 16 + fr][ks * 32 + fk * 8];
+          bf16x8 b = *(const bf16x8*)&lds_bt[nt * 16 + fr][ksl * 32 + fk * 8];
           acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
         }
       }
     }
+    __syncthreads();  // all waves done reading before next-step staging
   }
 
   // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + j
@@ -258,13 +258,13 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                         float* __restrict__ dWacc,
                         float* __restrict__ dBacc,  // [T, F] or nullptr
                         int T, int NB, int H, int W, int C,
-                        int Ho, int Wo, int F, int pad) {
+                        int Ho, int Wo, int F, int pad, int kchunk) {
   const int t = blockIdx.z;
   const int n0 = blockIdx.x * 64;          // column block within 9C
   const int N9 = 9 * C;
   const long Ktot = (long)NB * Ho * Wo;
-  const long kchunk0 = (long)blockIdx.y * WG_KCHUNK;
-  const long kchunk_end = min(kchunk0 + (long)WG_KCHUNK, Ktot);
+  const long kchunk0 = (long)blockIdx.y * kchunk;
+  const long kchunk_end = min(kchunk0 + (long)kchunk, Ktot);
   const int mtiles = (F + 15) / 16;
 
   __shared__ short lds_at[64][WBK + APAD];  // dY^T tile: [f][k]
@@ -513,15 +513,21 @@ std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
   auto acc = torch::zeros({T, N9, F}, x.options().dtype(torch::kFloat32));
   auto db = torch::zeros({T, F}, x.options().dtype(torch::kFloat32));
   const long Ktot = (long)NB * Ho * Wo;
-  dim3 grid((unsigned)((N9 + 63) / 64),
-            (unsigned)((Ktot + WG_KCHUNK - 1) / WG_KCHUNK), T);
+  // size K-chunks so the grid has >= ~1024 blocks (256 CUs want far more
+  // workgroups than CUs; small support-pass K was leaving the chip idle)
+  const int gridx = (N9 + 63) / 64;
+  long desired_y = std::max<long>(1, 1024 / std::max<long>(1, (long)gridx * T));
+  long kchunk = (Ktot + desired_y - 1) / desired_y;
+  kchunk = ((kchunk + WBK - 1) / WBK) * WBK;
+  kchunk = std::min<long>(std::max<long>(kchunk, WBK), WG_KCHUNK);
+  dim3 grid((unsigned)gridx, (unsigned)((Ktot + kchunk - 1) / kchunk), T);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(256), 0, stream.stream(),
                      reinterpret_cast<const bf16*>(dyc.data_ptr()),
                      reinterpret_cast<const bf16*>(xc.data_ptr()),
                      acc.data_ptr<float>(),
                      with_bias ? db.data_ptr<float>() : nullptr,
-                     T, NB, H, W, C, Ho, Wo, F, (int)pad);
+                     T, NB, H, W, C, Ho, Wo, F, (int)pad, (int)kchunk);
   auto dw = torch::empty({T, F, C, 3, 3}, x.options().dtype(torch::kFloat32));
   const long total = (long)T * F * C * 9;
   hipLaunchKernelGGL(wgrad_finalize_kernel, dim3(ew_grid2(total, 256)),
