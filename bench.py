@@ -33,7 +33,7 @@ def parse_cli():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--tasks_per_gpu", type=int, default=8)
+    p.add_argument("--tasks_per_gpu", type=int, default=32)
     p.add_argument("--inner_steps", type=int, default=5)
     p.add_argument("--model", type=str, default="maml++_miniimagenet_5w1s",
                    choices=["maml++_miniimagenet_5w1s", "maml++_omniglot_20w5s"])

@@ -159,6 +159,8 @@ def build_parser() -> argparse.ArgumentParser:
                    help="'auto' (nccl on GPU / gloo on CPU), 'nccl', or 'gloo'")
     p.add_argument("--synthetic_data", type=str, default="False",
                    help="use the synthetic episode stream (benchmarking; no dataset needed)")
+    p.add_argument("--enable_phase_timers", type=str, default="False",
+                   help="accumulate per-phase device timers (hipEvents)")
     return p
 
 
@@ -171,7 +173,7 @@ _BOOL_KEYS = (
     "enable_inner_loop_optimizable_bn_params", "second_order",
     "use_multi_step_loss_optimization", "per_step_bn_statistics",
     "learnable_batch_norm_momentum", "learnable_bn_gamma", "learnable_bn_beta",
-    "train_in_stages", "use_hip_kernels", "synthetic_data",
+    "train_in_stages", "use_hip_kernels", "synthetic_data", "enable_phase_timers",
 )
 
 

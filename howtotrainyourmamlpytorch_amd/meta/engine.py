@@ -79,6 +79,10 @@ class MAMLFewShotClassifier(nn.Module):
         self.optimizer = torch.optim.Adam(self.trainable_parameters(),
                                           lr=args.meta_learning_rate, amsgrad=False)
         self.dist = None  # set by attach_distributed()
+        self.timers = None
+        if getattr(args, "enable_phase_timers", False):
+            from ..utils.timers import PhaseTimers
+            self.timers = PhaseTimers(use_cuda_events=(device.type == "cuda"))
 
     # ------------------------------------------------------------------
     def attach_distributed(self, dist_ctx) -> None:
@@ -235,8 +239,14 @@ class MAMLFewShotClassifier(nn.Module):
         lr = self.scheduled_meta_lr(epoch)
         for group in self.optimizer.param_groups:
             group["lr"] = lr
-        losses, per_task_preds = self.train_forward_prop(data_batch, epoch)
-        self.meta_update(losses["loss"])
+        if self.timers is not None:
+            with self.timers.phase("inner_loop_fwd"):
+                losses, per_task_preds = self.train_forward_prop(data_batch, epoch)
+            with self.timers.phase("outer_bwd_and_opt"):
+                self.meta_update(losses["loss"])
+        else:
+            losses, per_task_preds = self.train_forward_prop(data_batch, epoch)
+            self.meta_update(losses["loss"])
         losses["loss"] = float(losses["loss"].detach().item())
         losses["learning_rate"] = lr
         self.current_iter += 1
