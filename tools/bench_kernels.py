@@ -11,7 +11,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 from howtotrainyourmamlpytorch_amd import ops  # noqa: E402
 from howtotrainyourmamlpytorch_amd.ops import hip_ext  # noqa: E402
 
