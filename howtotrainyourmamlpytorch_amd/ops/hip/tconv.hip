@@ -33,6 +33,15 @@ typedef __attribute__((__vector_size__(4 * sizeof(short)))) short short4v;
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
 
+// LDS XOR swizzle (guide T2 / G4): a [rows][64] bf16 tile has a 128-byte row
+// stride, so ds_read_b128 of 16 different rows at one column lands 8+ lanes
+// per bank.  XOR-ing row bits into the 16-byte-slot bits spreads them; the
+// same involution is applied on store and load.  Index in SHORTS; XOR of
+// bits 3..5 preserves 16-byte alignment for bf16x8 accesses.
+DEVINL int swz64(int row, int col) {
+  return (row * 64 + col) ^ ((row & 7) << 3);
+}
+
 // ---------------------------------------------------------------------------
 // Weight repack: W [T, F, C, 3, 3] fp32  ->  Wp [T, 9, Ci, Co] bf16
 //   fwd  : Wp[t][ky*3+kx][c][f] = W[t][f][c][ky][kx]          (Ci=C, Co=F)
@@ -103,8 +112,8 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
   const int ksteps = (K9 + BK - 1) / BK;
   const int ntiles = (Co + 15) / 16;
 
-  __shared__ short lds_a[BM][BK + APAD];
-  __shared__ short lds_bt[64][BK + APAD];   // B^T: [col][k]
+  __shared__ short lds_a[BM * BK];        // swizzled, see swz64
+  __shared__ short lds_bt[64 * BK];        // B^T: [col][k], swizzled
   __shared__ int row_h[BM], row_w[BM], row_n[BM];
   // full k-table hoisted out of the K-loop (9*Ci <= 576 always, since
   // Ci <= 64): one decode per k for the whole block, one fewer barrier
@@ -167,7 +176,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
                 (const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c0];
           }
         }
-        *(bf16x8*)&lds_a[m][kk0] = v;
+        *(bf16x8*)&lds_a[swz64(m, kk0)] = v;
       } else {
         for (int j = 0; j < 8; ++j) {
           const int kk = kk0 + j;
@@ -180,7 +189,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
               v = ((const short*)Xt)[(((long)n * H + h) * W + w) * Ci + c];
             }
           }
-          lds_a[m][kk] = v;
+          lds_a[swz64(m, kk & ~7) + (kk & 7)] = v;
         }
       }
     }
@@ -195,7 +204,8 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
           v = *(const bf16x8*)&((const short*)Wt)[(long)(k0 + kk) * Co + f0];
         }
 #pragma unroll
-        for (int j = 0; j < 8; ++j) lds_bt[f0 + j][kk] = v[j];
+        for (int j = 0; j < 8; ++j)
+          lds_bt[swz64(f0 + j, kk & ~7) + (kk & 7)] = v[j];
       } else {
         for (int j = 0; j < 8 && f0 + j < 64; ++j) {
           const int f = f0 + j;
@@ -203,7 +213,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
           if (f < Co && k0 + kk < K9) {
             v = ((const short*)Wt)[(long)(k0 + kk) * Co + f];
           }
-          lds_bt[f][kk] = v;
+          lds_bt[swz64(f, kk & ~7) + (kk & 7)] = v;
         }
       }
     }
@@ -212,11 +222,11 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     // fragments + MFMA (2 K-slices per staged tile)
 #pragma unroll
     for (int ksl = 0; ksl < BK / 32; ++ksl) {
-      bf16x8 a = *(const bf16x8*)&lds_a[wave * 16 + fr][ksl * 32 + fk * 8];
+      bf16x8 a = *(const bf16x8*)&lds_a[swz64(wave * 16 + fr, ksl * 32 + fk * 8)];
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         if (nt < ntiles) {
-          bf16x8 b = *(const bf16x8*)&lds_bt[nt * 16 + fr][ksl * 32 + fk * 8];
+          bf16x8 b = *(const bf16x8*)&lds_bt[swz64(nt * 16 + fr, ksl * 32 + fk * 8)];
           acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
         }
       }
@@ -267,8 +277,8 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   const long kchunk_end = min(kchunk0 + (long)kchunk, Ktot);
   const int mtiles = (F + 15) / 16;
 
-  __shared__ short lds_at[64][WBK + APAD];  // dY^T tile: [f][k]
-  __shared__ short lds_bt[64][WBK + APAD];  // im2col^T tile: [n][k]
+  __shared__ short lds_at[64 * WBK];  // dY^T tile: [f][k], swizzled
+  __shared__ short lds_bt[64 * WBK];  // im2col^T tile: [n][k], swizzled
   __shared__ int ntab_dy[64], ntab_dx[64], ntab_c[64];
   __shared__ int ktab_n[WBK], ktab_h[WBK], ktab_w[WBK];  // k -> image pos
 
@@ -332,7 +342,8 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
         v = *(const bf16x8*)&((const short*)dYt)[(k0 + kk) * F + f0];
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_at[f0 + j][kk] = v[j];
+      for (int j = 0; j < 8; ++j)
+        lds_at[swz64(f0 + j, kk & ~7) + (kk & 7)] = v[j];
     }
     // stage im2col^T: vector fast path when the 8-column run stays inside
     // one (ky,kx) slice (contiguous c, 16B-aligned when C % 8 == 0);
@@ -355,7 +366,8 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
           }
         }
 #pragma unroll
-        for (int j = 0; j < 8; ++j) lds_bt[n8 + j][kk] = v[j];
+        for (int j = 0; j < 8; ++j)
+          lds_bt[swz64(n8 + j, kk & ~7) + (kk & 7)] = v[j];
       } else {
         for (int j = 0; j < 8; ++j) {
           const int ncol = n8 + j;
@@ -368,7 +380,7 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
               v = ((const short*)Xt)[(((long)nimg * H + h) * W + w) * C + c];
             }
           }
-          lds_bt[ncol][kk] = v;
+          lds_bt[swz64(ncol, kk & ~7) + (kk & 7)] = v;
         }
       }
     }
@@ -376,11 +388,11 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
 
 #pragma unroll
     for (int ks = 0; ks < WBK / 32; ++ks) {
-      bf16x8 b = *(const bf16x8*)&lds_bt[wave * 16 + fr][ks * 32 + fk * 8];
+      bf16x8 b = *(const bf16x8*)&lds_bt[swz64(wave * 16 + fr, ks * 32 + fk * 8)];
 #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
         if (mt < mtiles) {
-          bf16x8 a = *(const bf16x8*)&lds_at[mt * 16 + fr][ks * 32 + fk * 8];
+          bf16x8 a = *(const bf16x8*)&lds_at[swz64(mt * 16 + fr, ks * 32 + fk * 8)];
           acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
         }
       }
@@ -388,8 +400,8 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     if (do_bias && db_f < F) {
 #pragma unroll
       for (int kk = db_q * 16; kk < db_q * 16 + 16; ++kk) {
-        db_acc += __bfloat162float(
-            __hip_bfloat16(__hip_bfloat16_raw{(unsigned short)lds_at[db_f][kk]}));
+        db_acc += __bfloat162float(__hip_bfloat16(__hip_bfloat16_raw{
+            (unsigned short)lds_at[swz64(db_f, kk & ~7) + (kk & 7)]}));
       }
     }
   }
