@@ -167,10 +167,19 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
 
     def _load_image(self, path: str) -> np.ndarray:
         """Decode + resize one image -> float32 HWC in [0, 1]
-        (reference: ``data.py:374-395``)."""
+        (reference: ``data.py:374-395``).  Corrupted files are tolerated
+        with a zero image + warning (the reference detects and re-saves,
+        ``data.py:280-300``; we cannot rewrite read-only datasets)."""
         from PIL import Image
 
-        img = Image.open(path)
+        try:
+            img = Image.open(path)
+            img.load()
+        except Exception:  # noqa: BLE001 - any decode failure
+            import warnings
+            warnings.warn(f"corrupted image {path!r}; substituting zeros")
+            return np.zeros((self.image_height, self.image_width,
+                             self.image_channels), dtype=np.float32)
         if self.image_channels == 1:
             img = img.convert("L")
         else:

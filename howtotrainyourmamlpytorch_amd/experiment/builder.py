@@ -148,12 +148,25 @@ class ExperimentBuilder:
             remaining = min(iters_per_epoch - (self.state["current_iter"] % iters_per_epoch),
                             total_iters - self.state["current_iter"])
             t0 = time.time()
+            pbar = None
+            if self.rank == 0:
+                try:
+                    from tqdm import tqdm
+                    pbar = tqdm(total=remaining, desc=f"epoch {epoch}", leave=False)
+                except ImportError:
+                    pass
             for train_sample in self.data.get_train_batches(
                     total_batches=remaining, augment_images=self.augment_flag):
                 epoch_idx = self.state["current_iter"] // iters_per_epoch
                 losses = self.train_iteration(train_sample, epoch_idx,
                                               self.state["current_iter"])
                 self.state["current_iter"] += 1
+                if pbar is not None:
+                    pbar.set_postfix(loss=f"{losses['loss']:.4f}",
+                                     acc=f"{losses['accuracy']:.4f}")
+                    pbar.update(1)
+            if pbar is not None:
+                pbar.close()
 
             if self.state["current_iter"] % iters_per_epoch == 0:
                 val_losses = []
