@@ -96,11 +96,11 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
 // pad is the im2col pad (fwd: p; dgrad: 2-p).  stride==1 only (stride-2
 // configs use the ATen fallback path).
 // ---------------------------------------------------------------------------
-#define BM 64
+#define BM 128
 #define BK 64    // K-step: 2 MFMA K-slices per barrier
 #define APAD 8   // bf16 row pad for LDS bank spread
 
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(512, 2)
 void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
                      const float* __restrict__ bias, bf16* __restrict__ Y,
                      int T, int NB, int H, int W, int Ci,
@@ -146,7 +146,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
   const bf16* Xt = X + (long)t * NB * H * W * Ci;
   const bf16* Wt = Wp + (long)t * 9 * Ci * Co;
 
-  const int wave = threadIdx.x / WAVE;          // 0..3: m-subtile
+  const int wave = threadIdx.x / WAVE;          // 0..7: m-subtile
   const int lane = threadIdx.x % WAVE;
   const int fr = lane & 15;                      // fragment row/col
   const int fk = lane >> 4;                      // k-group 0..3
@@ -263,27 +263,29 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 #define WG_KCHUNK 4096
 #define WBK 64   // wgrad K-step (2 MFMA K-slices per barrier)
 
-__global__ __launch_bounds__(256, 2)
+#define WGN 128  // wgrad output columns per block (8 waves x 16)
+
+__global__ __launch_bounds__(512, 2)
 void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                         float* __restrict__ dWacc,
                         float* __restrict__ dBacc,  // [T, F] or nullptr
                         int T, int NB, int H, int W, int C,
                         int Ho, int Wo, int F, int pad, int kchunk) {
   const int t = blockIdx.z;
-  const int n0 = blockIdx.x * 64;          // column block within 9C
+  const int n0 = blockIdx.x * WGN;         // column block within 9C
   const int N9 = 9 * C;
   const long Ktot = (long)NB * Ho * Wo;
   const long kchunk0 = (long)blockIdx.y * kchunk;
   const long kchunk_end = min(kchunk0 + (long)kchunk, Ktot);
   const int mtiles = (F + 15) / 16;
 
-  __shared__ short lds_at[64 * WBK];  // dY^T tile: [f][k], swizzled
-  __shared__ short lds_bt[64 * WBK];  // im2col^T tile: [n][k], swizzled
-  __shared__ int ntab_dy[64], ntab_dx[64], ntab_c[64];
+  __shared__ short lds_at[64 * WBK];   // dY^T tile: [f][k], swizzled
+  __shared__ short lds_bt[WGN * WBK];  // im2col^T tile: [n][k], swizzled
+  __shared__ int ntab_dy[WGN], ntab_dx[WGN], ntab_c[WGN];
   __shared__ int ktab_n[WBK], ktab_h[WBK], ktab_w[WBK];  // k -> image pos
 
   // n-table (once): n -> (dy, dx, c)
-  for (int e = threadIdx.x; e < 64; e += blockDim.x) {
+  for (int e = threadIdx.x; e < WGN; e += blockDim.x) {
     const int n = n0 + e;
     if (n < N9) {
       const int kyx = n / C;
@@ -311,7 +313,7 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   const bool do_bias = (dBacc != nullptr) && (blockIdx.x == 0);
   float db_acc = 0.f;
   const int db_f = threadIdx.x & 63;
-  const int db_q = threadIdx.x >> 6;        // quarter of the k range
+  const int db_q = threadIdx.x >> 6;        // k-range slice (8 x 8 with 512 thr)
 
   for (long k0 = kchunk0; k0 < kchunk_end; k0 += WBK) {
     __syncthreads();
@@ -348,9 +350,9 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     // stage im2col^T: vector fast path when the 8-column run stays inside
     // one (ky,kx) slice (contiguous c, 16B-aligned when C % 8 == 0);
     // scalar fallback otherwise (first-layer C in {1,3}).
-    for (int s = threadIdx.x; s < WBK * 8; s += blockDim.x) {
-      const int kk = s >> 3;
-      const int n8 = (s & 7) * 8;
+    for (int s = threadIdx.x; s < WBK * (WGN / 8); s += blockDim.x) {
+      const int kk = s / (WGN / 8);
+      const int n8 = (s % (WGN / 8)) * 8;
       const int nimg = ktab_n[kk];
       const int c0 = ntab_c[n8];
       const int c7 = ntab_c[n8 + 7];
@@ -399,20 +401,21 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     }
     if (do_bias && db_f < F) {
 #pragma unroll
-      for (int kk = db_q * 16; kk < db_q * 16 + 16; ++kk) {
+      for (int kk = db_q * 8; kk < db_q * 8 + 8; ++kk) {
         db_acc += __bfloat162float(__hip_bfloat16(__hip_bfloat16_raw{
             (unsigned short)lds_at[swz64(db_f, kk & ~7) + (kk & 7)]}));
       }
     }
   }
   if (do_bias) {
-    __shared__ float db_lds[4][64];
+    __shared__ float db_lds[8][64];
     db_lds[db_q][db_f] = db_acc;
     __syncthreads();
     if (db_q == 0 && db_f < F) {
-      atomicAdd(&dBacc[(long)t * F + db_f],
-                db_lds[0][db_f] + db_lds[1][db_f] + db_lds[2][db_f] +
-                    db_lds[3][db_f]);
+      float acc_b = 0.f;
+#pragma unroll
+      for (int q = 0; q < 8; ++q) acc_b += db_lds[q][db_f];
+      atomicAdd(&dBacc[(long)t * F + db_f], acc_b);
     }
   }
 
@@ -495,13 +498,14 @@ torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
   const long Mtot = (long)NB * Ho * Wo;
   dim3 grid((unsigned)((Mtot + BM - 1) / BM), T);
   auto stream = at::cuda::getCurrentCUDAStream();
+  const int mm_threads = 512;  // 8 waves: one 16-row m-subtile each
   const float* bptr = nullptr;
   torch::Tensor bc;
   if (bias.has_value()) {
     bc = bias->contiguous().to(torch::kFloat32);
     bptr = bc.data_ptr<float>();
   }
-  hipLaunchKernelGGL(tconv_mm_kernel, grid, dim3(256), 0, stream.stream(),
+  hipLaunchKernelGGL(tconv_mm_kernel, grid, dim3(mm_threads), 0, stream.stream(),
                      reinterpret_cast<const bf16*>(x.data_ptr()),
                      reinterpret_cast<const bf16*>(wp.data_ptr()), bptr,
                      reinterpret_cast<bf16*>(y.data_ptr()),
@@ -527,14 +531,14 @@ std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
   const long Ktot = (long)NB * Ho * Wo;
   // size K-chunks so the grid has >= ~1024 blocks (256 CUs want far more
   // workgroups than CUs; small support-pass K was leaving the chip idle)
-  const int gridx = (N9 + 63) / 64;
+  const int gridx = (N9 + WGN - 1) / WGN;
   long desired_y = std::max<long>(1, 1024 / std::max<long>(1, (long)gridx * T));
   long kchunk = (Ktot + desired_y - 1) / desired_y;
   kchunk = ((kchunk + WBK - 1) / WBK) * WBK;
   kchunk = std::min<long>(std::max<long>(kchunk, WBK), WG_KCHUNK);
   dim3 grid((unsigned)gridx, (unsigned)((Ktot + kchunk - 1) / kchunk), T);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(256), 0, stream.stream(),
+  hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(512), 0, stream.stream(),
                      reinterpret_cast<const bf16*>(dyc.data_ptr()),
                      reinterpret_cast<const bf16*>(xc.data_ptr()),
                      acc.data_ptr<float>(),
