@@ -263,9 +263,9 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 #define WG_KCHUNK 4096
 #define WBK 64   // wgrad K-step (2 MFMA K-slices per barrier)
 
-#define WGN 128  // wgrad output columns per block (8 waves x 16)
+#define WGN 64   // wgrad output columns per block (4 waves x 16)
 
-__global__ __launch_bounds__(512, 2)
+__global__ __launch_bounds__(256, 2)
 void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                         float* __restrict__ dWacc,
                         float* __restrict__ dBacc,  // [T, F] or nullptr
@@ -401,21 +401,20 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     }
     if (do_bias && db_f < F) {
 #pragma unroll
-      for (int kk = db_q * 8; kk < db_q * 8 + 8; ++kk) {
+      for (int kk = db_q * 16; kk < db_q * 16 + 16; ++kk) {
         db_acc += __bfloat162float(__hip_bfloat16(__hip_bfloat16_raw{
             (unsigned short)lds_at[swz64(db_f, kk & ~7) + (kk & 7)]}));
       }
     }
   }
   if (do_bias) {
-    __shared__ float db_lds[8][64];
+    __shared__ float db_lds[4][64];
     db_lds[db_q][db_f] = db_acc;
     __syncthreads();
     if (db_q == 0 && db_f < F) {
-      float acc_b = 0.f;
-#pragma unroll
-      for (int q = 0; q < 8; ++q) acc_b += db_lds[q][db_f];
-      atomicAdd(&dBacc[(long)t * F + db_f], acc_b);
+      atomicAdd(&dBacc[(long)t * F + db_f],
+                db_lds[0][db_f] + db_lds[1][db_f] + db_lds[2][db_f] +
+                    db_lds[3][db_f]);
     }
   }
 
@@ -538,7 +537,7 @@ std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
   kchunk = std::min<long>(std::max<long>(kchunk, WBK), WG_KCHUNK);
   dim3 grid((unsigned)gridx, (unsigned)((Ktot + kchunk - 1) / kchunk), T);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(512), 0, stream.stream(),
+  hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(256), 0, stream.stream(),
                      reinterpret_cast<const bf16*>(dyc.data_ptr()),
                      reinterpret_cast<const bf16*>(xc.data_ptr()),
                      acc.data_ptr<float>(),
