@@ -41,6 +41,9 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
 DEVINL int swz64(int row, int col) {
   return (row * 64 + col) ^ ((row & 7) << 3);
 }
+DEVINL int swz128(int row, int col) {
+  return (row * 128 + col) ^ ((row & 7) << 3);
+}
 
 // ---------------------------------------------------------------------------
 // Weight repack: W [T, F, C, 3, 3] fp32  ->  Wp [T, 9, Ci, Co] bf16
