@@ -140,6 +140,12 @@ class ExperimentBuilder:
 
     # ------------------------------------------------------------------
     def run_experiment(self) -> None:
+        if getattr(self.args, "evaluate_on_test_set_only", False):
+            # reference flag: skip training, evaluate the stored best
+            # checkpoints on the test set (experiment_builder.py:302-371)
+            self.evaluate_test_set_using_the_best_models(
+                top_n_models=getattr(self.args, "max_models_to_save", 5))
+            return
         total_iters = self.args.total_epochs * self.args.total_iter_per_epoch
         iters_per_epoch = self.args.total_iter_per_epoch
         epochs_done_this_run = 0
