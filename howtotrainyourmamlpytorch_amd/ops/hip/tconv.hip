@@ -161,11 +161,11 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
   __syncthreads();
   for (int ks = 0; ks < ksteps; ++ks) {
     const int k0 = ks * BK;
-    // stage A (im2col): slot s -> (m = s>>3, kk0 = (s&7)*8); vector path
-    // when the 8-k run stays inside one (ky,kx) slice (contiguous c).
-    for (int s = threadIdx.x; s < BM * 8; s += blockDim.x) {
-      const int m = s >> 3;
-      const int kk0 = (s & 7) * 8;
+    // stage A (im2col): slot s -> (m, kk0) with BK/8 slots per row;
+    // vector path when the 8-k run stays inside one (ky,kx) slice.
+    for (int s = threadIdx.x; s < BM * (BK / 8); s += blockDim.x) {
+      const int m = s / (BK / 8);
+      const int kk0 = (s % (BK / 8)) * 8;
       const int n = row_n[m];
       const int c0 = ktab_c[k0 + kk0];
       const int c7 = ktab_c[k0 + kk0 + 7];
@@ -199,7 +199,7 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     // stage B^T from the contiguous repacked block at linear k0*Co:
     // slot s -> (kk = s>>3, f0 = (s&7)*8), bf16x8 load + transposed writes
     for (int s = threadIdx.x; s < BK * 8; s += blockDim.x) {
-      const int kk = s >> 3;
+      const int kk = s >> 3;           // 8 f-slots per k row (Co <= 64)
       const int f0 = (s & 7) * 8;
       if ((Co % 8 == 0) && f0 + 8 <= Co) {
         bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
