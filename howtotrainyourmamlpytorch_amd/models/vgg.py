@@ -174,10 +174,18 @@ class TaskBatchedVGG(nn.Module):
             out = ops.task_conv3x3(out, w_i, b_i, stride=stride, padding=pad)
             if self.norm_layer_type == "batch_norm":
                 gamma, beta = self._bn_affine(i, num_step, v)
-                out, mean, var = ops.task_bn_act(out, gamma, beta, eps=self.bn_eps,
-                                                 negative_slope=self.negative_slope)
+                stat_count = out[0].numel() // out.shape[-1]
+                if self.max_pooling:
+                    # fused BN+act+pool (single pass over the conv output)
+                    out, mean, var = ops.task_bn_act_pool(
+                        out, gamma, beta, eps=self.bn_eps,
+                        negative_slope=self.negative_slope)
+                else:
+                    out, mean, var = ops.task_bn_act(
+                        out, gamma, beta, eps=self.bn_eps,
+                        negative_slope=self.negative_slope)
                 self._update_running_stats(i, num_step, mean, var,
-                                           count=out[0].numel() // out.shape[-1])
+                                           count=stat_count)
             elif self.norm_layer_type == "layer_norm":
                 wname = f"layer_dict.conv{i}.norm_layer.bias"
                 bias = v[wname] if wname in v else torch.zeros(
@@ -187,7 +195,7 @@ class TaskBatchedVGG(nn.Module):
                                               eps=self.bn_eps, negative_slope=self.negative_slope)
             else:
                 out = torch.nn.functional.leaky_relu(out, negative_slope=self.negative_slope)
-            if self.max_pooling:
+            if self.max_pooling and self.norm_layer_type != "batch_norm":
                 out = ops.task_maxpool2x2(out)
 
         if self.max_pooling:

@@ -97,6 +97,16 @@ def task_layer_norm_act(x, weight, bias, eps=1e-5, negative_slope=0.01, apply_ac
     return ref.task_layer_norm_act(x, weight, bias, eps, negative_slope, apply_act)
 
 
+def task_bn_act_pool(x, gamma, beta, eps=1e-5, negative_slope=0.01):
+    """BN(batch stats) + leaky-ReLU + 2x2 maxpool.  Fused single-pass
+    forward on GPU (C % 8 == 0); composition elsewhere."""
+    if x.is_cuda and x.shape[-1] % 8 == 0 and _want_hip(x):
+        from . import hip_autograd
+        return hip_autograd.task_bn_act_pool(x, gamma, beta, eps, negative_slope)
+    y, mean, var = task_bn_act(x, gamma, beta, eps, negative_slope)
+    return task_maxpool2x2(y), mean, var
+
+
 def task_maxpool2x2(x):
     if _want_hip(x):
         from . import hip_autograd

@@ -10,6 +10,9 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
                                       torch::Tensor mean, torch::Tensor rstd,
                                       torch::Tensor gamma, torch::Tensor beta,
                                       double slope, bool act);
+std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
+                                           torch::Tensor beta, double eps,
+                                           double slope);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor mask, long H, long W);
@@ -40,6 +43,7 @@ std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd, "fused task-batched BN+leakyReLU fwd");
   m.def("bn_act_bwd", &bn_act_bwd, "fused task-batched BN+leakyReLU bwd");
+  m.def("bn_act_pool_fwd", &bn_act_pool_fwd, "fused BN+leakyReLU+maxpool fwd");
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC maxpool 2x2 fwd");
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC maxpool 2x2 bwd");
   m.def("ce_fwd", &ce_fwd, "fused softmax-CE fwd");

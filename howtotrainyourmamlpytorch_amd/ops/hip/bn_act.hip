@@ -91,6 +91,67 @@ __global__ void bn_norm_act_vec_kernel(const scalar_t* __restrict__ x,
   }
 }
 
+// Fused normalize + leaky-ReLU + 2x2/s2 maxpool (forward only; the
+// backward composes the existing pool-bwd and BN-bwd Functions).  One
+// thread per (pooled position, 8-channel group): reads the 2x2 window of
+// conv output, normalizes all four, maxes, emits pooled value + argmax
+// mask.  Saves a full activation-tensor write+read per conv block.
+template <typename scalar_t, bool PER_TASK_AFFINE>
+__global__ void bn_norm_act_pool_vec_kernel(
+    const scalar_t* __restrict__ x,   // [T, NB, H, W, C]
+    scalar_t* __restrict__ y,         // [T, NB, Ho, Wo, C]
+    unsigned char* __restrict__ mask, // [T, NB, Ho, Wo, C]
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    int T, int NB, int H, int W, int C, int Ho, int Wo, float slope) {
+  const int c8n = C / 8;
+  const long total = (long)T * NB * Ho * Wo * c8n;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    const int c8 = (int)(i % c8n);
+    long r = i / c8n;
+    const int wo = (int)(r % Wo); r /= Wo;
+    const int ho = (int)(r % Ho); r /= Ho;
+    const int nb = (int)(r % NB);
+    const int t = (int)(r / NB);
+    const int c0 = c8 * 8;
+    float mu[8], rs[8], g[8], b[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const long tc = (long)t * C + c0 + j;
+      mu[j] = mean[tc];
+      rs[j] = rstd[tc];
+      g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c0 + j];
+      b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c0 + j];
+    }
+    const long base =
+        ((((long)t * NB + nb) * H + 2 * ho) * W + 2 * wo) * C + c0;
+    float v[4][8];
+    load8(&x[base], v[0]);
+    load8(&x[base + C], v[1]);
+    load8(&x[base + (long)W * C], v[2]);
+    load8(&x[base + (long)W * C + C], v[3]);
+    unsigned long long mpack = 0;
+    float out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float best = -INFINITY;
+      int arg = 0;
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        float a = (v[q][j] - mu[j]) * rs[j] * g[j] + b[j];
+        a = a > 0.f ? a : a * slope;
+        if (a > best) { best = a; arg = q; }
+      }
+      out[j] = best;
+      mpack |= ((unsigned long long)arg) << (8 * j);
+    }
+    const long o = ((((long)t * NB + nb) * Ho + ho) * Wo + wo) * C + c0;
+    store8(&y[o], out);
+    *(unsigned long long*)&mask[o] = mpack;
+  }
+}
+
 template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
 __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
                                        const scalar_t* __restrict__ x,
@@ -571,4 +632,66 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
   // dbeta = bsums[:,0,:], dgamma = bsums[:,1,:]  (per task; wrapper reduces
   // over T when gamma is shared)
   return {dx, bsums.select(1, 1).clone(), bsums.select(1, 0).clone()};
+}
+
+// x: [T, NB, H, W, C] contiguous bf16/fp32; returns
+// {y_pooled [T, NB, H/2, W/2, C], mask u8, mean, var, rstd}
+// Fuses normalize+leakyReLU+maxpool into one pass over the conv output;
+// requires C % 8 == 0 (callers fall back to the unfused ops otherwise).
+std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
+                                           torch::Tensor beta, double eps,
+                                           double slope) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 5 && x.is_contiguous());
+  const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3), C = (int)x.size(4);
+  TORCH_CHECK(C % 8 == 0 && C <= 512, "bn_act_pool_fwd needs C % 8 == 0");
+  const long M = (long)NB * H * W;
+  const int Ho = H / 2, Wo = W / 2;
+  auto fopts = x.options().dtype(torch::kFloat32);
+  auto sums = torch::zeros({T, 2, C}, fopts);
+  auto mean = torch::empty({T, C}, fopts);
+  auto var = torch::empty({T, C}, fopts);
+  auto rstd = torch::empty({T, C}, fopts);
+  auto y = torch::empty({T, NB, Ho, Wo, C}, x.options());
+  auto mask = torch::empty({T, NB, Ho, Wo, C}, x.options().dtype(torch::kUInt8));
+  auto gc = gamma.contiguous().to(torch::kFloat32);
+  auto bc = beta.contiguous().to(torch::kFloat32);
+  const bool per_task = gamma.dim() == 2;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int rpb = 1024;
+  dim3 gsums(T, (unsigned)((M + rpb - 1) / rpb));
+  const long ptotal = (long)T * NB * Ho * Wo * (C / 8);
+  const int pblocks = (int)std::min<long>((ptotal + 255) / 256, 4096);
+
+#define LAUNCH_BNP(ST, PT)                                                     \
+  do {                                                                         \
+    hipLaunchKernelGGL((bn_sums_vec_kernel<ST>), gsums, dim3(256),             \
+                       2 * C * (int)sizeof(float), stream.stream(),            \
+                       reinterpret_cast<const ST*>(x.data_ptr()),              \
+                       sums.data_ptr<float>(), T, M, C, rpb);                  \
+    const int fin_blocks = (T * C + 255) / 256;                                \
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(fin_blocks), dim3(256), 0,     \
+                       stream.stream(), sums.data_ptr<float>(),                \
+                       mean.data_ptr<float>(), var.data_ptr<float>(),          \
+                       rstd.data_ptr<float>(), T, M, C, (float)eps);           \
+    hipLaunchKernelGGL((bn_norm_act_pool_vec_kernel<ST, PT>), dim3(pblocks),   \
+                       dim3(256), 0, stream.stream(),                          \
+                       reinterpret_cast<const ST*>(x.data_ptr()),              \
+                       reinterpret_cast<ST*>(y.data_ptr()),                    \
+                       mask.data_ptr<unsigned char>(),                         \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
+                       gc.data_ptr<float>(), bc.data_ptr<float>(),             \
+                       T, NB, H, W, C, Ho, Wo, (float)slope);                  \
+  } while (0)
+
+  if (x.scalar_type() == torch::kFloat32) {
+    if (per_task) LAUNCH_BNP(float, true); else LAUNCH_BNP(float, false);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    if (per_task) LAUNCH_BNP(__hip_bfloat16, true);
+    else LAUNCH_BNP(__hip_bfloat16, false);
+  } else {
+    TORCH_CHECK(false, "bn_act_pool_fwd: unsupported dtype");
+  }
+#undef LAUNCH_BNP
+  return {y, mask, mean, var, rstd};
 }

@@ -98,6 +98,45 @@ def task_bn_act(x, gamma, beta, eps=1e-5, negative_slope=0.01, apply_act=True):
     return y.view(T, NS, H, W, C), mean, var
 
 
+class _BNActPoolFn(torch.autograd.Function):
+    """Fused normalize+leakyReLU+2x2 maxpool FORWARD (one pass over the
+    conv output instead of write+reread of the activation); the backward
+    composes the existing differentiable pool-bwd and BN-bwd Functions, so
+    second-order behavior is identical to the unfused pair."""
+
+    @staticmethod
+    def forward(ctx, x5, gamma, beta, eps, slope):
+        y, mask, mean, var, rstd = _ext().bn_act_pool_fwd(
+            x5, gamma.float(), beta.float(), eps, slope)
+        ctx.save_for_backward(x5, gamma, beta, mean, rstd, mask)
+        ctx.slope = slope
+        ctx.mark_non_differentiable(mean, var)
+        return y, mean, var
+
+    @staticmethod
+    def backward(ctx, dy, dmean, dvar):
+        x5, gamma, beta, mean, rstd, mask = ctx.saved_tensors
+        T, NB, H, W, C = x5.shape
+        Ho, Wo = H // 2, W // 2
+        da = _PoolBwdFn.apply(dy.contiguous().view(T * NB, Ho, Wo, C),
+                              mask.view(T * NB, Ho, Wo, C), H, W)
+        dx, dgamma_t, dbeta_t = _BNBwdFn.apply(
+            x5.view(T, NB * H * W, C), gamma, beta,
+            da.view(T, NB * H * W, C), mean, rstd, ctx.slope, True)
+        if gamma.dim() == 2:
+            dgamma, dbeta = dgamma_t, dbeta_t
+        else:
+            dgamma, dbeta = dgamma_t.sum(0), dbeta_t.sum(0)
+        return (dx.view(T, NB, H, W, C), dgamma.to(gamma.dtype),
+                dbeta.to(beta.dtype), None, None)
+
+
+def task_bn_act_pool(x, gamma, beta, eps=1e-5, negative_slope=0.01):
+    y, mean, var = _BNActPoolFn.apply(x.contiguous(), gamma, beta, eps,
+                                      negative_slope)
+    return y, mean, var
+
+
 # ---------------------------------------------------------------------------
 # maxpool 2x2 — mask is fixed after forward, so backward (scatter) and
 # double-backward (gather) are both linear custom kernels.

@@ -212,3 +212,36 @@ def test_engine_gpu_bf16_kernels_close_to_cpu():
     # end-to-end); fp32 tight equivalence is asserted separately above
     assert rel_l2 < 0.40, f"relative L2 {rel_l2:.4f}"
     assert cos > 0.95, f"cosine {cos:.5f}"
+
+
+def test_bn_act_pool_fused_matches_composition():
+    torch.manual_seed(7)
+    T, NS, H, W, C = 3, 5, 10, 10, 48
+    x = torch.randn(T, NS, H, W, C, device=dev(), requires_grad=True)
+    gamma = (torch.rand(C, device=dev()) + 0.5).requires_grad_(True)
+    beta = torch.randn(C, device=dev()).requires_grad_(True)
+
+    y_f, mean_f, var_f = ops.task_bn_act_pool(x, gamma, beta)
+    y_c0, mean_c, var_c = ops.task_bn_act(x, gamma, beta)
+    y_c = ops.task_maxpool2x2(y_c0)
+    torch.testing.assert_close(y_f, y_c, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(mean_f, mean_c, rtol=1e-5, atol=1e-6)
+
+    g = torch.randn_like(y_f)
+    gf = torch.autograd.grad(y_f, (x, gamma, beta), g, retain_graph=True)
+    gc = torch.autograd.grad(y_c, (x, gamma, beta), g, retain_graph=True)
+    for a, b in zip(gf, gc):
+        torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-4)
+
+    # second-order through the fused op matches the composition
+    def grad_norm(y_, x_):
+        (gx,) = torch.autograd.grad((y_.float() ** 2).mean(), (x_,), create_graph=True)
+        return (gx.float() ** 2).sum()
+
+    x2 = x.detach().requires_grad_(True)
+    y2, _, _ = ops.task_bn_act_pool(x2, gamma.detach(), beta.detach())
+    (gg_f,) = torch.autograd.grad(grad_norm(y2, x2), (x2,))
+    x3 = x.detach().requires_grad_(True)
+    y3 = ops.task_maxpool2x2(ops.task_bn_act(x3, gamma.detach(), beta.detach())[0])
+    (gg_c,) = torch.autograd.grad(grad_norm(y3, x3), (x3,))
+    torch.testing.assert_close(gg_f, gg_c, rtol=1e-3, atol=1e-4)
