@@ -171,15 +171,21 @@ class TaskBatchedVGG(nn.Module):
         for i in range(self.num_stages):
             w_i = v[f"layer_dict.conv{i}.conv.weight"]
             b_i = v[f"layer_dict.conv{i}.conv.bias"]
-            out = ops.task_conv3x3(out, w_i, b_i, stride=stride, padding=pad)
+            want_stats = (self.norm_layer_type == "batch_norm"
+                          and self.max_pooling)
+            if want_stats:
+                out, bn_sums = ops.task_conv3x3(out, w_i, b_i, stride=stride,
+                                                padding=pad, return_stats=True)
+            else:
+                out = ops.task_conv3x3(out, w_i, b_i, stride=stride, padding=pad)
             if self.norm_layer_type == "batch_norm":
                 gamma, beta = self._bn_affine(i, num_step, v)
                 stat_count = out[0].numel() // out.shape[-1]
                 if self.max_pooling:
-                    # fused BN+act+pool (single pass over the conv output)
+                    # fused BN+act+pool consuming the conv-epilogue stats
                     out, mean, var = ops.task_bn_act_pool(
                         out, gamma, beta, eps=self.bn_eps,
-                        negative_slope=self.negative_slope)
+                        negative_slope=self.negative_slope, sums=bn_sums)
                 else:
                     out, mean, var = ops.task_bn_act(
                         out, gamma, beta, eps=self.bn_eps,

@@ -78,11 +78,12 @@ def _want_hip(x: torch.Tensor) -> bool:
 # public ops — HIP autograd wrappers are registered here as they land
 # ---------------------------------------------------------------------------
 
-def task_conv3x3(x, w, b=None, stride=1, padding=1):
+def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
     if _want_hip(x):
         from . import hip_autograd
-        return hip_autograd.task_conv3x3(x, w, b, stride, padding)
-    return ref.task_conv3x3(x, w, b, stride, padding)
+        return hip_autograd.task_conv3x3(x, w, b, stride, padding, return_stats)
+    y = ref.task_conv3x3(x, w, b, stride, padding)
+    return (y, None) if return_stats else y
 
 
 def task_bn_act(x, gamma, beta, eps=1e-5, negative_slope=0.01, apply_act=True):
@@ -97,12 +98,14 @@ def task_layer_norm_act(x, weight, bias, eps=1e-5, negative_slope=0.01, apply_ac
     return ref.task_layer_norm_act(x, weight, bias, eps, negative_slope, apply_act)
 
 
-def task_bn_act_pool(x, gamma, beta, eps=1e-5, negative_slope=0.01):
+def task_bn_act_pool(x, gamma, beta, eps=1e-5, negative_slope=0.01, sums=None):
     """BN(batch stats) + leaky-ReLU + 2x2 maxpool.  Fused single-pass
-    forward on GPU (C % 8 == 0); composition elsewhere."""
+    forward on GPU (C % 8 == 0), optionally consuming conv-epilogue
+    precomputed stats; composition elsewhere."""
     if x.is_cuda and x.shape[-1] % 8 == 0 and _want_hip(x):
         from . import hip_autograd
-        return hip_autograd.task_bn_act_pool(x, gamma, beta, eps, negative_slope)
+        return hip_autograd.task_bn_act_pool(x, gamma, beta, eps,
+                                             negative_slope, sums)
     y, mean, var = task_bn_act(x, gamma, beta, eps, negative_slope)
     return task_maxpool2x2(y), mean, var
 

@@ -12,7 +12,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
                                       double slope, bool act);
 std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
                                            torch::Tensor beta, double eps,
-                                           double slope);
+                                           double slope,
+                                           c10::optional<torch::Tensor> sums_in);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor mask, long H, long W);
@@ -33,9 +34,9 @@ std::vector<torch::Tensor> lslr_bwd(torch::Tensor gout, torch::Tensor grad,
                                     torch::Tensor lr_vec);
 // tconv.hip
 torch::Tensor tconv_repack(torch::Tensor w, bool dgrad);
-torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
-                       c10::optional<torch::Tensor> bias, long pad,
-                       long Ho, long Wo);
+std::vector<torch::Tensor> tconv_mm(torch::Tensor x, torch::Tensor wp,
+                                    c10::optional<torch::Tensor> bias, long pad,
+                                    long Ho, long Wo, bool with_stats);
 std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
                                        long pad, bool with_bias);
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B);

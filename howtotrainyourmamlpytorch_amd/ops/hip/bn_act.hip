@@ -640,7 +640,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
 // requires C % 8 == 0 (callers fall back to the unfused ops otherwise).
 std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
                                            torch::Tensor beta, double eps,
-                                           double slope) {
+                                           double slope,
+                                           c10::optional<torch::Tensor> sums_in) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 5 && x.is_contiguous());
   const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
             W = (int)x.size(3), C = (int)x.size(4);
@@ -648,7 +649,9 @@ std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
   const long M = (long)NB * H * W;
   const int Ho = H / 2, Wo = W / 2;
   auto fopts = x.options().dtype(torch::kFloat32);
-  auto sums = torch::zeros({T, 2, C}, fopts);
+  const bool have_sums = sums_in.has_value() && sums_in->numel() == T * 2 * C;
+  auto sums = have_sums ? sums_in->contiguous()
+                        : torch::zeros({T, 2, C}, fopts);
   auto mean = torch::empty({T, C}, fopts);
   auto var = torch::empty({T, C}, fopts);
   auto rstd = torch::empty({T, C}, fopts);
@@ -665,10 +668,12 @@ std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
 
 #define LAUNCH_BNP(ST, PT)                                                     \
   do {                                                                         \
-    hipLaunchKernelGGL((bn_sums_vec_kernel<ST>), gsums, dim3(256),             \
-                       2 * C * (int)sizeof(float), stream.stream(),            \
-                       reinterpret_cast<const ST*>(x.data_ptr()),              \
-                       sums.data_ptr<float>(), T, M, C, rpb);                  \
+    if (!have_sums) {                                                          \
+      hipLaunchKernelGGL((bn_sums_vec_kernel<ST>), gsums, dim3(256),           \
+                         2 * C * (int)sizeof(float), stream.stream(),          \
+                         reinterpret_cast<const ST*>(x.data_ptr()),            \
+                         sums.data_ptr<float>(), T, M, C, rpb);                \
+    }                                                                          \
     const int fin_blocks = (T * C + 255) / 256;                                \
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(fin_blocks), dim3(256), 0,     \
                        stream.stream(), sums.data_ptr<float>(),                \

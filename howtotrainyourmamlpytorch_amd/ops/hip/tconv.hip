@@ -106,7 +106,8 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
 __global__ __launch_bounds__(512, 2)
 void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
                      const float* __restrict__ bias, bf16* __restrict__ Y,
-                     int T, int NB, int H, int W, int Ci,
+                     float* __restrict__ sums_out,  // [T,2,Co] or nullptr:
+                     int T, int NB, int H, int W, int Ci,   // fused BN stats
                      int Ho, int Wo, int Co, int pad) {
   const int t = blockIdx.y;
   const long Mtot = (long)NB * Ho * Wo;
@@ -237,7 +238,17 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     __syncthreads();  // all waves done reading before next-step staging
   }
 
-  // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + j
+  // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + j.
+  // When sums_out is given, per-column sum/sum-of-squares of the outputs
+  // (pre-bf16-rounding, bias included) accumulate for the following BN —
+  // removing BN's separate stats pass over the conv output.
+  __shared__ float sums_lds[2][64];
+  if (sums_out) {
+    for (int i = threadIdx.x; i < 128; i += blockDim.x) {
+      sums_lds[i >> 6][i & 63] = 0.f;
+    }
+    __syncthreads();
+  }
   bf16* Yt = Y + (long)t * Mtot * Co;
 #pragma unroll
   for (int nt = 0; nt < 4; ++nt) {
@@ -245,14 +256,29 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     const int col = nt * 16 + fr;
     if (col >= Co) continue;
     const float bv = bias ? bias[(long)t * Co + col] : 0.f;
+    float ls = 0.f, lq = 0.f;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const int row = wave * 16 + fk * 4 + j;
       const long mg = m0 + row;
       if (mg < Mtot) {
+        const float v = acc[nt][j] + bv;
         ((short*)Yt)[mg * Co + col] =
-            (short)__bfloat16_as_short(__float2bfloat16(acc[nt][j] + bv));
+            (short)__bfloat16_as_short(__float2bfloat16(v));
+        ls += v;
+        lq += v * v;
       }
+    }
+    if (sums_out) {
+      atomicAdd(&sums_lds[0][col], ls);
+      atomicAdd(&sums_lds[1][col], lq);
+    }
+  }
+  if (sums_out) {
+    __syncthreads();
+    for (int c = threadIdx.x; c < Co; c += blockDim.x) {
+      atomicAdd(&sums_out[((long)t * 2 + 0) * Co + c], sums_lds[0][c]);
+      atomicAdd(&sums_out[((long)t * 2 + 1) * Co + c], sums_lds[1][c]);
     }
   }
 }
@@ -487,9 +513,10 @@ torch::Tensor tconv_repack(torch::Tensor w, bool dgrad) {
 }
 
 // x [T, NB, H, W, Ci] bf16 ; wp [T, 9, Ci, Co] bf16 ; bias [T, Co] fp32 / undef
-torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
-                       c10::optional<torch::Tensor> bias, long pad,
-                       long Ho, long Wo) {
+// returns {y, sums[T,2,Co]} — sums populated only when with_stats
+std::vector<torch::Tensor> tconv_mm(torch::Tensor x, torch::Tensor wp,
+                                    c10::optional<torch::Tensor> bias, long pad,
+                                    long Ho, long Wo, bool with_stats) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 5 && x.is_contiguous());
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "tconv_mm needs bf16");
   const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
@@ -497,6 +524,9 @@ torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
   const int Co = (int)wp.size(3);
   TORCH_CHECK(wp.size(2) == Ci && Co <= 64, "Co must be <= 64");
   auto y = torch::empty({T, NB, Ho, Wo, Co}, x.options());
+  auto sums = with_stats
+                  ? torch::zeros({T, 2, Co}, x.options().dtype(torch::kFloat32))
+                  : torch::empty({0}, x.options().dtype(torch::kFloat32));
   const long Mtot = (long)NB * Ho * Wo;
   dim3 grid((unsigned)((Mtot + BM - 1) / BM), T);
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -511,8 +541,9 @@ torch::Tensor tconv_mm(torch::Tensor x, torch::Tensor wp,
                      reinterpret_cast<const bf16*>(x.data_ptr()),
                      reinterpret_cast<const bf16*>(wp.data_ptr()), bptr,
                      reinterpret_cast<bf16*>(y.data_ptr()),
+                     with_stats ? sums.data_ptr<float>() : nullptr,
                      T, NB, H, W, Ci, (int)Ho, (int)Wo, Co, (int)pad);
-  return y;
+  return {y, sums};
 }
 
 // dy [T, NB, Ho, Wo, F] bf16 ; x [T, NB, H, W, C] bf16

@@ -105,9 +105,9 @@ class _BNActPoolFn(torch.autograd.Function):
     second-order behavior is identical to the unfused pair."""
 
     @staticmethod
-    def forward(ctx, x5, gamma, beta, eps, slope):
+    def forward(ctx, x5, gamma, beta, eps, slope, sums):
         y, mask, mean, var, rstd = _ext().bn_act_pool_fwd(
-            x5, gamma.float(), beta.float(), eps, slope)
+            x5, gamma.float(), beta.float(), eps, slope, sums)
         ctx.save_for_backward(x5, gamma, beta, mean, rstd, mask)
         ctx.slope = slope
         ctx.mark_non_differentiable(mean, var)
@@ -128,12 +128,12 @@ class _BNActPoolFn(torch.autograd.Function):
         else:
             dgamma, dbeta = dgamma_t.sum(0), dbeta_t.sum(0)
         return (dx.view(T, NB, H, W, C), dgamma.to(gamma.dtype),
-                dbeta.to(beta.dtype), None, None)
+                dbeta.to(beta.dtype), None, None, None)
 
 
-def task_bn_act_pool(x, gamma, beta, eps=1e-5, negative_slope=0.01):
+def task_bn_act_pool(x, gamma, beta, eps=1e-5, negative_slope=0.01, sums=None):
     y, mean, var = _BNActPoolFn.apply(x.contiguous(), gamma, beta, eps,
-                                      negative_slope)
+                                      negative_slope, sums)
     return y, mean, var
 
 
@@ -256,18 +256,23 @@ def lslr_update(arena, grad, lr_vec):
 # create_graph included), no torch fallback on the hot path.
 # ---------------------------------------------------------------------------
 class _ConvFwdFn(torch.autograd.Function):
+    """Returns (y, bn_sums): with want_stats the epilogue accumulates the
+    following BN's per-channel sum/sum-of-squares for free."""
+
     @staticmethod
-    def forward(ctx, x, w, b, pad):
+    def forward(ctx, x, w, b, pad, want_stats):
         ctx.save_for_backward(x, w)
         ctx.pad = pad
         ctx.has_bias = b is not None
         wp = _ext().tconv_repack(w, False)
         H, W = x.shape[2], x.shape[3]
         Ho, Wo = H + 2 * pad - 2, W + 2 * pad - 2
-        return _ext().tconv_mm(x, wp, b, pad, Ho, Wo)
+        y, sums = _ext().tconv_mm(x, wp, b, pad, Ho, Wo, want_stats)
+        ctx.mark_non_differentiable(sums)
+        return y, sums
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, dsums):
         x, w = ctx.saved_tensors
         dy = dy.contiguous()
         dx = dw = db = None
@@ -280,7 +285,11 @@ class _ConvFwdFn(torch.autograd.Function):
                 db = None
         elif ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(1, 2, 3))
-        return dx, dw, db, None
+        return dx, dw, db, None, None
+
+
+def _conv_fwd(x, w, b, pad):
+    return _ConvFwdFn.apply(x, w, b, pad, False)[0]
 
 
 class _ConvDgradFn(torch.autograd.Function):
@@ -294,7 +303,7 @@ class _ConvDgradFn(torch.autograd.Function):
         wp = _ext().tconv_repack(w, True)
         Ho, Wo = dy.shape[2], dy.shape[3]
         H, W = Ho - 2 * pad + 2, Wo - 2 * pad + 2
-        return _ext().tconv_mm(dy, wp, None, 2 - pad, H, W)
+        return _ext().tconv_mm(dy, wp, None, 2 - pad, H, W, False)[0]
 
     @staticmethod
     def backward(ctx, g):
@@ -302,7 +311,7 @@ class _ConvDgradFn(torch.autograd.Function):
         g = g.contiguous()
         d_dy = d_w = None
         if ctx.needs_input_grad[0]:
-            d_dy = _ConvFwdFn.apply(g, w, None, ctx.pad)
+            d_dy = _conv_fwd(g, w, None, ctx.pad)
         if ctx.needs_input_grad[1]:
             d_w = _ConvWgradFn.apply(dy, g, ctx.pad, False)[0]
         return d_dy, d_w, None
@@ -326,7 +335,7 @@ class _ConvWgradFn(torch.autograd.Function):
         d_dy = d_x = None
         if ctx.needs_input_grad[0]:
             if gw is not None:
-                d_dy = _ConvFwdFn.apply(x, gw.contiguous(), None, ctx.pad)
+                d_dy = _conv_fwd(x, gw.contiguous(), None, ctx.pad)
             if gdb is not None:
                 T, F = gdb.shape
                 add = gdb.view(T, 1, 1, 1, F).to(dy.dtype)
@@ -336,15 +345,18 @@ class _ConvWgradFn(torch.autograd.Function):
         return d_dy, d_x, None, None
 
 
-def task_conv3x3(x, w, b=None, stride=1, padding=1):
+def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
     if stride != 1 or x.dtype != torch.bfloat16 or w.shape[1] > 64:
         # stride-2 (max_pooling=False configs) and fp32 compute use the
         # grouped-ATen composition; cast for dtype consistency
         wc = w.to(x.dtype)
         bc = b.to(x.dtype) if b is not None else None
-        return ref.task_conv3x3(x, wc, bc, stride, padding)
-    return _ConvFwdFn.apply(x.contiguous(), w.contiguous(),
-                            b.contiguous() if b is not None else None, padding)
+        y = ref.task_conv3x3(x, wc, bc, stride, padding)
+        return (y, None) if return_stats else y
+    y, sums = _ConvFwdFn.apply(x.contiguous(), w.contiguous(),
+                               b.contiguous() if b is not None else None,
+                               padding, return_stats)
+    return (y, sums) if return_stats else y
 
 
 def task_linear(x, w, b=None):
