@@ -14,6 +14,10 @@ std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
                                            torch::Tensor beta, double eps,
                                            double slope,
                                            c10::optional<torch::Tensor> sums_in);
+std::vector<torch::Tensor> bn_act_pool_bwd(torch::Tensor dyp, torch::Tensor mask,
+                                           torch::Tensor x, torch::Tensor mean,
+                                           torch::Tensor rstd, torch::Tensor gamma,
+                                           torch::Tensor beta, double slope);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor mask, long H, long W);
@@ -45,6 +49,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd, "fused task-batched BN+leakyReLU fwd");
   m.def("bn_act_bwd", &bn_act_bwd, "fused task-batched BN+leakyReLU bwd");
   m.def("bn_act_pool_fwd", &bn_act_pool_fwd, "fused BN+leakyReLU+maxpool fwd");
+  m.def("bn_act_pool_bwd", &bn_act_pool_bwd, "fused BN+leakyReLU+maxpool bwd");
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC maxpool 2x2 fwd");
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC maxpool 2x2 bwd");
   m.def("ce_fwd", &ce_fwd, "fused softmax-CE fwd");

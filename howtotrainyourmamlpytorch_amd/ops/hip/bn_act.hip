@@ -351,6 +351,149 @@ __global__ void bn_norm_act_kernel(const scalar_t* __restrict__ x,
   }
 }
 
+// --------------------------------------------------------------------------
+// Fused backward of BN+act+maxpool (used when the backward pass itself is
+// not being differentiated — outer backward / eval; the create_graph path
+// composes the individual Functions instead).  The pooled incoming grad is
+// expanded through the argmax mask inline, so the full-resolution `da`
+// tensor is never materialized.
+//   da(t,nb,h,w,c) = mask match ? dyp(t,nb,h/2,w/2,c) : 0
+// then the standard BN+act backward math on da.
+// --------------------------------------------------------------------------
+template <typename scalar_t, bool PER_TASK_AFFINE>
+DEVINL void pool_expand8(const scalar_t* __restrict__ dyp,
+                         const unsigned char* __restrict__ mask,
+                         long row, int NB, int H, int W, int C,
+                         int c0, float* dv) {
+  const int Ho = H / 2, Wo = W / 2;
+  const int w = (int)(row % W);
+  const int h = (int)((row / W) % H);
+  const long nb = row / ((long)W * H);
+  const int ho = h >> 1, wo = w >> 1;
+  if (ho < Ho && wo < Wo) {
+    const long o = ((nb * Ho + ho) * Wo + wo) * C + c0;
+    const int arg = ((h & 1) << 1) | (w & 1);
+    const unsigned long long mp = *(const unsigned long long*)&mask[o];
+    float dvv[8];
+    load8(&dyp[o], dvv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      dv[j] = ((int)((mp >> (8 * j)) & 0xff) == arg) ? dvv[j] : 0.f;
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dv[j] = 0.f;
+  }
+}
+
+template <typename scalar_t, bool PER_TASK_AFFINE>
+__global__ void bn_pool_bwd_sums_vec_kernel(
+    const scalar_t* __restrict__ dyp,      // [T, NB, Ho, Wo, C]
+    const unsigned char* __restrict__ mask,
+    const scalar_t* __restrict__ x,        // [T, NB, H, W, C]
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ bsums,             // [T, 2, C]
+    int T, int NB, int H, int W, int C, float slope, int rows_per_block) {
+  const long M = (long)NB * H * W;
+  const int c8n = C / 8;
+  const int rows_in_block = blockDim.x / c8n;
+  const int c8 = threadIdx.x % c8n;
+  const int rg = threadIdx.x / c8n;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  extern __shared__ float ls[];
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
+  __syncthreads();
+  if (rg < rows_in_block) {
+    const int c0 = c8 * 8;
+    float mu[8], r[8], g[8], b[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const long tc = (long)t * C + c0 + j;
+      mu[j] = mean[tc]; r[j] = rstd[tc];
+      g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c0 + j];
+      b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c0 + j];
+    }
+    float s1[8] = {0}, s2[8] = {0};
+    const scalar_t* xt = x + (long)t * M * C + c0;
+    const scalar_t* dypt = dyp + (long)t * NB * (H / 2) * (W / 2) * C;
+    const unsigned char* mt_ = mask + (long)t * NB * (H / 2) * (W / 2) * C;
+    const long row_end = min(row0 + rows_per_block, M);
+    for (long m = row0 + rg; m < row_end; m += rows_in_block) {
+      float xv[8], dv[8];
+      load8(xt + m * C, xv);
+      pool_expand8<scalar_t, PER_TASK_AFFINE>(dypt, mt_, m, NB, H, W, C, c0, dv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xh = (xv[j] - mu[j]) * r[j];
+        float d = dv[j];
+        d *= ((xh * g[j] + b[j]) > 0.f) ? 1.f : slope;
+        s1[j] += d;
+        s2[j] += d * xh;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&ls[c0 + j], s1[j]);
+      atomicAdd(&ls[C + c0 + j], s2[j]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    atomicAdd(&bsums[((long)t * 2 + 0) * C + i], ls[i]);
+    atomicAdd(&bsums[((long)t * 2 + 1) * C + i], ls[C + i]);
+  }
+}
+
+template <typename scalar_t, bool PER_TASK_AFFINE>
+__global__ void bn_pool_bwd_dx_vec_kernel(
+    const scalar_t* __restrict__ dyp, const unsigned char* __restrict__ mask,
+    const scalar_t* __restrict__ x, scalar_t* __restrict__ dx,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ bsums,
+    int T, int NB, int H, int W, int C, float slope, int rows_per_block) {
+  const long M = (long)NB * H * W;
+  const int c8n = C / 8;
+  const int rows_in_block = blockDim.x / c8n;
+  const int c8 = threadIdx.x % c8n;
+  const int rg = threadIdx.x / c8n;
+  if (rg >= rows_in_block) return;
+  const int t = blockIdx.x;
+  const long row0 = (long)blockIdx.y * rows_per_block;
+  const float invM = 1.f / (float)M;
+  const int c0 = c8 * 8;
+  float mu[8], r[8], g[8], b[8], a1[8], a2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const long tc = (long)t * C + c0 + j;
+    mu[j] = mean[tc]; r[j] = rstd[tc];
+    g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c0 + j];
+    b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c0 + j];
+    a1[j] = bsums[((long)t * 2 + 0) * C + c0 + j] * invM;
+    a2[j] = bsums[((long)t * 2 + 1) * C + c0 + j] * invM;
+  }
+  const scalar_t* xt = x + (long)t * M * C + c0;
+  scalar_t* dxt = dx + (long)t * M * C + c0;
+  const scalar_t* dypt = dyp + (long)t * NB * (H / 2) * (W / 2) * C;
+  const unsigned char* mt_ = mask + (long)t * NB * (H / 2) * (W / 2) * C;
+  const long row_end = min(row0 + rows_per_block, M);
+  for (long m = row0 + rg; m < row_end; m += rows_in_block) {
+    float xv[8], dv[8];
+    load8(xt + m * C, xv);
+    pool_expand8<scalar_t, PER_TASK_AFFINE>(dypt, mt_, m, NB, H, W, C, c0, dv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float xh = (xv[j] - mu[j]) * r[j];
+      float d = dv[j];
+      d *= ((xh * g[j] + b[j]) > 0.f) ? 1.f : slope;
+      xv[j] = g[j] * r[j] * (d - a1[j] - xh * a2[j]);
+    }
+    store8(dxt + m * C, xv);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Backward stage 1: per-(t,c) sums  s1 = sum dy', s2 = sum dy' * xhat,
 // where dy' = dy * act'(pre-act) and xhat = (x - mean) * rstd.
@@ -699,4 +842,62 @@ std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
   }
 #undef LAUNCH_BNP
   return {y, mask, mean, var, rstd};
+}
+
+// Fused backward of BN+act+pool: dyp [T,NB,Ho,Wo,C], mask u8, x [T,NB,H,W,C]
+// -> {dx [T,NB,H,W,C], dgamma_t [T,C], dbeta_t [T,C]}
+std::vector<torch::Tensor> bn_act_pool_bwd(torch::Tensor dyp, torch::Tensor mask,
+                                           torch::Tensor x, torch::Tensor mean,
+                                           torch::Tensor rstd, torch::Tensor gamma,
+                                           torch::Tensor beta, double slope) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 5 && x.is_contiguous());
+  const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3), C = (int)x.size(4);
+  TORCH_CHECK(C % 8 == 0 && C <= 512);
+  const long M = (long)NB * H * W;
+  auto fopts = x.options().dtype(torch::kFloat32);
+  auto bsums = torch::zeros({T, 2, C}, fopts);
+  auto dx = torch::empty_like(x);
+  auto gc = gamma.contiguous().to(torch::kFloat32);
+  auto bc = beta.contiguous().to(torch::kFloat32);
+  auto dypc = dyp.contiguous();
+  const bool per_task = gamma.dim() == 2;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int rpb = 1024;
+  dim3 g(T, (unsigned)((M + rpb - 1) / rpb));
+
+#define LAUNCH_BPB(ST, PT)                                                     \
+  do {                                                                         \
+    hipLaunchKernelGGL((bn_pool_bwd_sums_vec_kernel<ST, PT>), g, dim3(256),    \
+                       2 * C * (int)sizeof(float), stream.stream(),            \
+                       reinterpret_cast<const ST*>(dypc.data_ptr()),           \
+                       mask.data_ptr<unsigned char>(),                         \
+                       reinterpret_cast<const ST*>(x.data_ptr()),              \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
+                       gc.data_ptr<float>(), bc.data_ptr<float>(),             \
+                       bsums.data_ptr<float>(), T, NB, H, W, C, (float)slope,  \
+                       rpb);                                                   \
+    hipLaunchKernelGGL((bn_pool_bwd_dx_vec_kernel<ST, PT>), g, dim3(256), 0,   \
+                       stream.stream(),                                        \
+                       reinterpret_cast<const ST*>(dypc.data_ptr()),           \
+                       mask.data_ptr<unsigned char>(),                         \
+                       reinterpret_cast<const ST*>(x.data_ptr()),              \
+                       reinterpret_cast<ST*>(dx.data_ptr()),                   \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
+                       gc.data_ptr<float>(), bc.data_ptr<float>(),             \
+                       bsums.data_ptr<float>(), T, NB, H, W, C, (float)slope,  \
+                       rpb);                                                   \
+  } while (0)
+
+  if (x.scalar_type() == torch::kFloat32) {
+    if (per_task) LAUNCH_BPB(float, true); else LAUNCH_BPB(float, false);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    if (per_task) LAUNCH_BPB(__hip_bfloat16, true);
+    else LAUNCH_BPB(__hip_bfloat16, false);
+  } else {
+    TORCH_CHECK(false, "bn_act_pool_bwd: unsupported dtype");
+  }
+#undef LAUNCH_BPB
+  // dbeta_t = bsums[:,0], dgamma_t = bsums[:,1]
+  return {dx, bsums.select(1, 1).clone(), bsums.select(1, 0).clone()};
 }

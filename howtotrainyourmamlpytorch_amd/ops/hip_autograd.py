@@ -118,16 +118,25 @@ class _BNActPoolFn(torch.autograd.Function):
         x5, gamma, beta, mean, rstd, mask = ctx.saved_tensors
         T, NB, H, W, C = x5.shape
         Ho, Wo = H // 2, W // 2
-        da = _PoolBwdFn.apply(dy.contiguous().view(T * NB, Ho, Wo, C),
-                              mask.view(T * NB, Ho, Wo, C), H, W)
-        dx, dgamma_t, dbeta_t = _BNBwdFn.apply(
-            x5.view(T, NB * H * W, C), gamma, beta,
-            da.view(T, NB * H * W, C), mean, rstd, ctx.slope, True)
+        if torch.is_grad_enabled():
+            # create_graph (second-order inner loop): compose the
+            # differentiable Functions
+            da = _PoolBwdFn.apply(dy.contiguous().view(T * NB, Ho, Wo, C),
+                                  mask.view(T * NB, Ho, Wo, C), H, W)
+            dx, dgamma_t, dbeta_t = _BNBwdFn.apply(
+                x5.view(T, NB * H * W, C), gamma, beta,
+                da.view(T, NB * H * W, C), mean, rstd, ctx.slope, True)
+            dx = dx.view(T, NB, H, W, C)
+        else:
+            # plain backward: fully fused (no da materialization)
+            dx, dgamma_t, dbeta_t = _ext().bn_act_pool_bwd(
+                dy.contiguous(), mask, x5, mean, rstd, gamma.float(),
+                beta.float(), ctx.slope)
         if gamma.dim() == 2:
             dgamma, dbeta = dgamma_t, dbeta_t
         else:
             dgamma, dbeta = dgamma_t.sum(0), dbeta_t.sum(0)
-        return (dx.view(T, NB, H, W, C), dgamma.to(gamma.dtype),
+        return (dx, dgamma.to(gamma.dtype),
                 dbeta.to(beta.dtype), None, None, None)
 
 
