@@ -171,8 +171,10 @@ class TaskBatchedVGG(nn.Module):
         for i in range(self.num_stages):
             w_i = v[f"layer_dict.conv{i}.conv.weight"]
             b_i = v[f"layer_dict.conv{i}.conv.bias"]
+            import os
             want_stats = (self.norm_layer_type == "batch_norm"
-                          and self.max_pooling)
+                          and self.max_pooling
+                          and os.environ.get("MAML355_NO_EPIFUSE", "0") != "1")
             if want_stats:
                 out, bn_sums = ops.task_conv3x3(out, w_i, b_i, stride=stride,
                                                 padding=pad, return_stats=True)
