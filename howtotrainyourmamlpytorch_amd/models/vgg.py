@@ -175,6 +175,7 @@ class TaskBatchedVGG(nn.Module):
             want_stats = (self.norm_layer_type == "batch_norm"
                           and self.max_pooling
                           and os.environ.get("MAML355_NO_EPIFUSE", "0") != "1")
+            bn_sums = None
             if want_stats:
                 out, bn_sums = ops.task_conv3x3(out, w_i, b_i, stride=stride,
                                                 padding=pad, return_stats=True)
