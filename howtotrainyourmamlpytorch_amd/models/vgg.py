@@ -171,10 +171,13 @@ class TaskBatchedVGG(nn.Module):
         for i in range(self.num_stages):
             w_i = v[f"layer_dict.conv{i}.conv.weight"]
             b_i = v[f"layer_dict.conv{i}.conv.bias"]
+            # conv-epilogue BN-stats fusion measured NET-NEGATIVE (-9%:
+            # LDS atomics in the hot conv kernel cost more than the cheap
+            # separate stats pass) — keep the capability but default off
             import os
             want_stats = (self.norm_layer_type == "batch_norm"
                           and self.max_pooling
-                          and os.environ.get("MAML355_NO_EPIFUSE", "0") != "1")
+                          and os.environ.get("MAML355_EPIFUSE", "0") == "1")
             bn_sums = None
             if want_stats:
                 out, bn_sums = ops.task_conv3x3(out, w_i, b_i, stride=stride,
