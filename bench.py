@@ -4,7 +4,8 @@ Config = BASELINE.json's headline: Mini-ImageNet 5-way 1-shot MAML++
 (48 filters, 84x84x3, 5 inner steps, LSLR+MSL, second-order), synthetic
 episodes, random-init weights, bf16 conv compute / fp32 master.
 
-Weak scaling: per-GPU meta-batch is fixed (default 8 tasks/GPU); the
+Weak scaling: per-GPU meta-batch is fixed (default 128 tasks/GPU, sized
+well within 288 GB HBM3E per BASELINE.json config #5); the
 reported ``value`` is whole-job tasks/sec over all ranks.
 
 Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
@@ -33,7 +34,7 @@ def parse_cli():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--tasks_per_gpu", type=int, default=32)
+    p.add_argument("--tasks_per_gpu", type=int, default=128)
     p.add_argument("--inner_steps", type=int, default=5)
     p.add_argument("--model", type=str, default="maml++_miniimagenet_5w1s",
                    choices=["maml++_miniimagenet_5w1s", "maml++_omniglot_20w5s"])
