@@ -19,7 +19,7 @@ Differences from the reference that matter:
 
 from __future__ import annotations
 
-import math
+import os
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -174,7 +174,6 @@ class TaskBatchedVGG(nn.Module):
             # conv-epilogue BN-stats fusion measured NET-NEGATIVE (-9%:
             # LDS atomics in the hot conv kernel cost more than the cheap
             # separate stats pass) — keep the capability but default off
-            import os
             want_stats = (self.norm_layer_type == "batch_norm"
                           and self.max_pooling
                           and os.environ.get("MAML355_EPIFUSE", "0") == "1")

@@ -16,11 +16,14 @@
 // accumulate via v_mfma_f32_16x16x32_bf16.  Layout: NHWC (channel-
 // innermost = K-contiguous im2col rows).
 //
-// Geometry (fwd/dgrad): block = 256 threads = 4 waves; BM=64 output rows,
-// BN = Cout (<= 64, i.e. up to 4 n-tiles of 16), BK=32.  Wave w owns
-// 16-row m-subtile w and iterates all n-tiles.  A-tile staged by im2col
-// gather (LDS, +8 bf16 row pad), B-tile contiguous from the repacked
-// weights, transposed into LDS for K-contiguous fragment reads.
+// Geometry (fwd/dgrad): block = 512 threads = 8 waves; BM=128 output rows,
+// BN = Cout (<= 64, i.e. up to 4 n-tiles of 16), BK=64 (2 MFMA K-slices
+// per barrier).  Wave w owns 16-row m-subtile w and iterates all n-tiles.
+// A-tile staged by im2col gather, B-tile contiguous from the repacked
+// weights, transposed into LDS for K-contiguous fragment reads; both
+// tiles XOR-swizzled (swz64).  Geometry settled by same-box A/B sweeps —
+// see profiles/README.md (BM 64->128 +50% TF; BK=128 and 2-half-wave
+// variants measured slower via LDS-occupancy / issue-pressure cliffs).
 
 #include "common.h"
 #include <torch/extension.h>
@@ -29,7 +32,6 @@
 using namespace maml355;
 
 using bf16 = __hip_bfloat16;
-typedef __attribute__((__vector_size__(4 * sizeof(short)))) short short4v;
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
 
@@ -101,7 +103,6 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
 // ---------------------------------------------------------------------------
 #define BM 128
 #define BK 64    // K-step: 2 MFMA K-slices per barrier
-#define APAD 8   // bf16 row pad for LDS bank spread
 
 __global__ __launch_bounds__(512, 2)
 void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,

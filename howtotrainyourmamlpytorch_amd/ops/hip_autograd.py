@@ -3,21 +3,25 @@
 Double-backward policy (second-order MAML needs ``create_graph=True``
 through the inner-loop support forward):
 
-* ops whose backward is *bilinear in saved tensors* (LSLR update, maxpool
-  scatter) use custom kernels at every order;
-* BN+act and softmax-CE check ``torch.is_grad_enabled()`` inside
-  ``backward`` (True exactly under create_graph) and fall back to a
-  differentiable torch composition there, using the fused backward kernels
-  on every ordinary backward (eval, first-order, and the big outer
-  backward).
+* the conv trio (fwd/dgrad/wgrad) is mutually bilinear — each backward
+  composes the other two kernels, so convs run on custom kernels at every
+  derivative order;
+* BN+act and softmax-CE make their *first backward* its own Function
+  whose backward evaluates the analytic closed-form second derivative
+  (bn_dbwd.hip / ce_dbwd) — also custom kernels at every order;
+* LSLR update and maxpool are bilinear in their saved tensors — custom at
+  every order;
+* the fused BN+act+pool op uses a single-pass fused backward when the
+  backward itself is not being differentiated, and composes the
+  individual Functions under create_graph.
 
-Conv uses the fully-custom bilinear trio in ``tconv.hip`` once built; until
-then the grouped-conv reference composition runs (MIOpen under ATen).
+Stride-2 convs (max_pooling=False configs) and fp32 compute fall back to
+the grouped-ATen composition.
 """
 
 from __future__ import annotations
 
-from typing import Optional
+import os
 
 import torch
 
@@ -118,7 +122,6 @@ class _BNActPoolFn(torch.autograd.Function):
         x5, gamma, beta, mean, rstd, mask = ctx.saved_tensors
         T, NB, H, W, C = x5.shape
         Ho, Wo = H // 2, W // 2
-        import os
         if torch.is_grad_enabled() or os.environ.get("MAML355_NO_BWDFUSE", "0") == "1":
             # create_graph (second-order inner loop): compose the
             # differentiable Functions
