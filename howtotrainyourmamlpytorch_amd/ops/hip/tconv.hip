@@ -101,7 +101,7 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
 // pad is the im2col pad (fwd: p; dgrad: 2-p).  stride==1 only (stride-2
 // configs use the ATen fallback path).
 // ---------------------------------------------------------------------------
-#define BM 128
+#define BM 256
 #define BK 64    // K-step: 2 MFMA K-slices per barrier
 
 __global__ __launch_bounds__(512, 2)
@@ -151,14 +151,16 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
   const bf16* Xt = X + (long)t * NB * H * W * Ci;
   const bf16* Wt = Wp + (long)t * 9 * Ci * Co;
 
-  const int wave = threadIdx.x / WAVE;          // 0..7: m-subtile
+  const int wave = threadIdx.x / WAVE;          // 0..7: 32-row m-subtile
   const int lane = threadIdx.x % WAVE;
   const int fr = lane & 15;                      // fragment row/col
   const int fk = lane >> 4;                      // k-group 0..3
 
-  f32x4 acc[4];
+  f32x4 acc[2][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[hh][i] = {0.f, 0.f, 0.f, 0.f};
 
   __syncthreads();
   for (int ks = 0; ks < ksteps; ++ks) {
@@ -224,15 +226,18 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     }
     __syncthreads();
 
-    // fragments + MFMA (2 K-slices per staged tile)
+    // fragments + MFMA (2 K-slices, 2 row-halves per wave: each B frag
+    // feeds two MFMAs)
 #pragma unroll
     for (int ksl = 0; ksl < BK / 32; ++ksl) {
-      bf16x8 a = *(const bf16x8*)&lds_a[swz64(wave * 16 + fr, ksl * 32 + fk * 8)];
+      bf16x8 a0 = *(const bf16x8*)&lds_a[swz64(wave * 32 + fr, ksl * 32 + fk * 8)];
+      bf16x8 a1 = *(const bf16x8*)&lds_a[swz64(wave * 32 + 16 + fr, ksl * 32 + fk * 8)];
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         if (nt < ntiles) {
           bf16x8 b = *(const bf16x8*)&lds_bt[swz64(nt * 16 + fr, ksl * 32 + fk * 8)];
-          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+          acc[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc[0][nt], 0, 0, 0);
+          acc[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, acc[1][nt], 0, 0, 0);
         }
       }
     }
@@ -252,6 +257,8 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
   }
   bf16* Yt = Y + (long)t * Mtot * Co;
 #pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
   for (int nt = 0; nt < 4; ++nt) {
     if (nt >= ntiles) break;
     const int col = nt * 16 + fr;
@@ -260,10 +267,10 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
     float ls = 0.f, lq = 0.f;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const int row = wave * 16 + fk * 4 + j;
+      const int row = wave * 32 + hh * 16 + fk * 4 + j;
       const long mg = m0 + row;
       if (mg < Mtot) {
-        const float v = acc[nt][j] + bv;
+        const float v = acc[hh][nt][j] + bv;
         ((short*)Yt)[mg * Co + col] =
             (short)__bfloat16_as_short(__float2bfloat16(v));
         ls += v;
