@@ -161,6 +161,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="use the synthetic episode stream (benchmarking; no dataset needed)")
     p.add_argument("--enable_phase_timers", type=str, default="False",
                    help="accumulate per-phase device timers (hipEvents)")
+    p.add_argument("--task_chunk_size", type=int, default=0,
+                   help="0 = whole local meta-batch in one graph; >0 = "
+                        "accumulate outer grads over task chunks of this "
+                        "size (caps the second-order activation tape)")
     return p
 
 

@@ -230,6 +230,41 @@ class MAMLFewShotClassifier(nn.Module):
                     p.grad.data.clamp_(-10, 10)
         self.optimizer.step()
 
+    def _chunked_train_step(self, data_batch, epoch: int, chunk: int):
+        """Per-chunk forward+backward with gradient accumulation.
+
+        The outer loss is a mean over tasks, so running ``chunk`` tasks at a
+        time and scaling each chunk's backward by ``chunk_size/total`` is
+        mathematically identical to the single big graph (SURVEY.md §7
+        "memory of the unrolled tape") — it caps the live second-order
+        activation tape to one chunk, letting the meta-batch exceed what a
+        single unrolled graph fits in HBM."""
+        total = data_batch[0].shape[0]
+        self.optimizer.zero_grad(set_to_none=True)
+        agg_loss = 0.0
+        agg_acc = 0.0
+        preds = []
+        base_losses = None
+        for lo in range(0, total, chunk):
+            hi = min(lo + chunk, total)
+            sub = tuple(x[lo:hi] for x in data_batch)
+            losses, p = self.train_forward_prop(sub, epoch)
+            (losses["loss"] * ((hi - lo) / total)).backward()
+            agg_loss += float(losses["loss"].detach()) * (hi - lo) / total
+            agg_acc += losses["accuracy"] * (hi - lo) / total
+            preds.append(p)
+            base_losses = losses
+        if self.dist is not None and self.dist.world_size > 1:
+            self.dist.all_reduce_gradients(self.trainable_parameters())
+        if "imagenet" in self.args.dataset_name:
+            for p_ in self.trainable_parameters():
+                if p_.grad is not None:
+                    p_.grad.data.clamp_(-10, 10)
+        self.optimizer.step()
+        base_losses["loss"] = agg_loss
+        base_losses["accuracy"] = agg_acc
+        return base_losses, torch.cat(preds, dim=0)
+
     def run_train_iter(self, data_batch, epoch):
         epoch = int(epoch)
         if self.current_epoch != epoch:
@@ -239,6 +274,12 @@ class MAMLFewShotClassifier(nn.Module):
         lr = self.scheduled_meta_lr(epoch)
         for group in self.optimizer.param_groups:
             group["lr"] = lr
+        chunk = int(getattr(self.args, "task_chunk_size", 0) or 0)
+        if chunk > 0 and chunk < data_batch[0].shape[0]:
+            losses, per_task_preds = self._chunked_train_step(data_batch, epoch, chunk)
+            losses["learning_rate"] = lr
+            self.current_iter += 1
+            return losses, per_task_preds
         if self.timers is not None:
             with self.timers.phase("inner_loop_fwd"):
                 losses, per_task_preds = self.train_forward_prop(data_batch, epoch)
