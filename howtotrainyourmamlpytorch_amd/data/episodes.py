@@ -24,7 +24,7 @@ from __future__ import annotations
 import concurrent.futures
 import json
 import os
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Sequence
 
 import numpy as np
 import torch

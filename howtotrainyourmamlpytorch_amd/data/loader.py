@@ -12,7 +12,7 @@ floating-point reduction order.
 
 from __future__ import annotations
 
-from typing import Iterator, Optional
+from typing import Iterator
 
 import torch
 

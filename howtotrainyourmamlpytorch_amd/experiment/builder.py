@@ -22,8 +22,7 @@ from typing import Dict, List, Optional
 import numpy as np
 import torch
 
-from .storage import (build_experiment_folder, load_from_json, save_statistics,
-                      save_to_json)
+from .storage import build_experiment_folder, save_statistics, save_to_json
 
 
 class ExperimentBuilder:

@@ -12,7 +12,7 @@ from __future__ import annotations
 import csv
 import json
 import os
-from typing import Any, Dict, Iterable, List, Sequence, Tuple
+from typing import Any, Dict, List, Sequence, Tuple
 
 
 def build_experiment_folder(experiment_name: str, root: str = ".") -> Tuple[str, str, str]:

@@ -20,7 +20,7 @@ Layers see the arena through zero-copy views produced by :meth:`views`.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Dict, Iterable, List, Sequence, Tuple
+from typing import Dict, List, Sequence, Tuple
 
 import torch
 

@@ -16,7 +16,6 @@ invalidate every benchmark, so we raise instead.
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 import torch
 

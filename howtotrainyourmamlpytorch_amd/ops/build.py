@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import os
 import shutil
-import sys
 
 PKG_DIR = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(PKG_DIR, "hip")
