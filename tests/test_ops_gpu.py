@@ -245,3 +245,21 @@ def test_bn_act_pool_fused_matches_composition():
     y3 = ops.task_maxpool2x2(ops.task_bn_act(x3, gamma.detach(), beta.detach())[0])
     (gg_c,) = torch.autograd.grad(grad_norm(y3, x3), (x3,))
     torch.testing.assert_close(gg_f, gg_c, rtol=1e-3, atol=1e-4)
+
+
+def test_bn_act_pool_fused_per_task_affine():
+    """inner-loop-optimizable BN params: per-task gamma/beta through the
+    fused BN+act+pool path."""
+    torch.manual_seed(8)
+    T, NS, H, W, C = 2, 3, 8, 8, 48
+    x = torch.randn(T, NS, H, W, C, device=dev(), requires_grad=True)
+    gamma = (torch.rand(T, C, device=dev()) + 0.5).requires_grad_(True)
+    beta = torch.randn(T, C, device=dev()).requires_grad_(True)
+    y_f, _, _ = ops.task_bn_act_pool(x, gamma, beta)
+    y_c = ops.task_maxpool2x2(ops.task_bn_act(x, gamma, beta)[0])
+    torch.testing.assert_close(y_f, y_c, rtol=1e-4, atol=1e-4)
+    g = torch.randn_like(y_f)
+    gf = torch.autograd.grad(y_f, (x, gamma, beta), g, retain_graph=True)
+    gc = torch.autograd.grad(y_c, (x, gamma, beta), g)
+    for a, b in zip(gf, gc):
+        torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-4)
