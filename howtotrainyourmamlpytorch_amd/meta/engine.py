@@ -161,9 +161,11 @@ class MAMLFewShotClassifier(nn.Module):
                                              training=True,
                                              backup_running_statistics=(step == 0))
             support_loss = ops.softmax_cross_entropy(support_logits, ys)  # [T]
+            # first-order / eval: the support graph is not traversed again,
+            # so its activation buffers are freed here (retain only under
+            # create_graph, where it is implied)
             grad = torch.autograd.grad(support_loss.sum(), arena,
-                                       create_graph=use_second_order,
-                                       retain_graph=True)[0]
+                                       create_graph=use_second_order)[0]
             lr_vec = self.classifier.lr_vector(self.inner_loop_lrs, step)
             arena = ops.lslr_update(arena, grad, lr_vec)
 
@@ -173,9 +175,13 @@ class MAMLFewShotClassifier(nn.Module):
                 if step == num_steps - 1:
                     final_logits = tl
             elif step == num_steps - 1:
-                final_logits = self.classifier(xt, num_step=step, arena=arena,
-                                               training=True)
-                per_step_target_loss.append(ops.softmax_cross_entropy(final_logits, yt))
+                # eval / final-step-only: target loss is report-only unless
+                # training — skip graph construction at eval
+                with torch.set_grad_enabled(training_phase):
+                    final_logits = self.classifier(xt, num_step=step, arena=arena,
+                                                   training=True)
+                    per_step_target_loss.append(
+                        ops.softmax_cross_entropy(final_logits, yt))
 
         if msl_active:
             step_losses = torch.stack(per_step_target_loss, dim=0)        # [steps, T]
