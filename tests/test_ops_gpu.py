@@ -263,3 +263,30 @@ def test_bn_act_pool_fused_per_task_affine():
     gc = torch.autograd.grad(y_c, (x, gamma, beta), g)
     for a, b in zip(gf, gc):
         torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-4)
+
+
+def test_bn_act_double_backward_per_task_affine():
+    """Second-order through BN with inner-loop-adapted (per-task) gamma/beta
+    — the PER_TASK_AFFINE template path of bn_dbwd."""
+    torch.manual_seed(9)
+    T, NS, H, W, C = 2, 3, 4, 4, 48
+    x = torch.randn(T, NS, H, W, C, device=dev(), requires_grad=True)
+    gamma = (torch.rand(T, C, device=dev()) + 0.5).requires_grad_(True)
+    beta = torch.randn(T, C, device=dev()).requires_grad_(True)
+
+    def loss_of_grad(op, x_, g_, b_):
+        y, _, _ = op(x_, g_, b_)
+        l = (y.float() ** 2).mean()
+        gx, gg = torch.autograd.grad(l, (x_, g_), create_graph=True)
+        return (gx.float() ** 2).sum() + (gg.float() ** 2).sum()
+
+    l2 = loss_of_grad(ops.task_bn_act, x, gamma, beta)
+    gg = torch.autograd.grad(l2, (x, gamma))
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    gr = gamma.detach().float().cpu().requires_grad_(True)
+    br = beta.detach().float().cpu().requires_grad_(True)
+    l2r = loss_of_grad(ref.task_bn_act, xr, gr, br)
+    ggr = torch.autograd.grad(l2r, (xr, gr))
+    for a, b in zip(gg, ggr):
+        torch.testing.assert_close(a.cpu(), b, rtol=2e-3, atol=2e-3)
