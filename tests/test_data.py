@@ -135,3 +135,26 @@ def test_synthetic_stream_shapes_and_sharding():
     b = next(iter(r1.get_train_batches(1)))
     merged = torch.cat([a[0], b[0]], dim=0)
     torch.testing.assert_close(merged, xs)
+
+
+def test_pre_split_dataset_layout(tmp_path):
+    """sets_are_pre_split: train/ val/ test/ subdirectories with their own
+    class folders (the mini-imagenet layout, reference data.py:234-268)."""
+    from PIL import Image
+    rng = np.random.RandomState(1)
+    root = tmp_path / "presplit_ds"
+    counts = {"train": 5, "val": 3, "test": 3}
+    for split, ncls in counts.items():
+        for c in range(ncls):
+            d = root / split / f"group{c}" / f"cls{c}"
+            d.mkdir(parents=True)
+            for i in range(4):
+                arr = rng.randint(0, 255, size=(10, 10), dtype=np.uint8)
+                Image.fromarray(arr, mode="L").save(d / f"{i}.png")
+    args = data_args(str(root), sets_are_pre_split=True)
+    ds = FewShotEpisodeDataset(args, current_set="train")
+    assert len(ds.datasets["train"]) == 5
+    assert len(ds.datasets["val"]) == 3
+    assert len(ds.datasets["test"]) == 3
+    xs, xt, ys, yt, seed = ds.get_set("val", seed=3)
+    assert xs.shape == (3, 2, 1, 10, 10)
