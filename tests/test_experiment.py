@@ -114,3 +114,36 @@ def test_from_scratch_ignores_existing_checkpoint(tmp_path):
     args2 = exp_args(tmp_path, name="scratch_exp", continue_from_epoch="from_scratch")
     builder2 = build(args2)
     assert builder2.state["current_iter"] == 0
+
+
+def test_full_experiment_on_real_omniglot_npz(tmp_path):
+    """Builder + real episode loader (shipped npz) + checkpointing +
+    ensemble test, end to end on CPU with a tiny model."""
+    args = get_args([
+        "--experiment_name", "omni_cpu_exp",
+        "--experiment_root", str(tmp_path),
+        "--dataset_name", "omniglot_dataset",
+        "--dataset_path", "datasets/omniglot_28x28.npz",
+        "--batch_size", "2",
+        "--num_classes_per_set", "3",
+        "--num_samples_per_class", "1",
+        "--num_target_samples", "1",
+        "--image_height", "28", "--image_width", "28", "--image_channels", "1",
+        "--cnn_num_filters", "4", "--num_stages", "3",
+        "--number_of_training_steps_per_iter", "1",
+        "--number_of_evaluation_steps_per_iter", "1",
+        "--total_epochs", "1", "--total_iter_per_epoch", "2",
+        "--num_evaluation_tasks", "2",
+        "--max_models_to_save", "1",
+        "--num_dataprovider_workers", "0",
+        "--seed", "2",
+    ])
+    from howtotrainyourmamlpytorch_amd.data import MetaLearningSystemDataLoader
+    device = torch.device("cpu")
+    model = MAMLFewShotClassifier(im_shape=(2, 1, 28, 28), device=device, args=args)
+    data = MetaLearningSystemDataLoader(args)
+    builder = ExperimentBuilder(args=args, data=data, model=model, device=device)
+    builder.run_experiment()
+    base = tmp_path / "omni_cpu_exp"
+    assert (base / "saved_models" / "train_model_latest").is_file()
+    assert (base / "logs" / "test_summary.csv").is_file()
