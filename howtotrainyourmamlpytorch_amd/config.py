@@ -143,6 +143,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--learnable_bn_beta", type=str, default="True")
     p.add_argument("--meta_learning_rate", type=float, default=0.001)
     p.add_argument("--min_learning_rate", type=float, default=0.00001)
+    # weight_decay is accepted and intentionally dead, exactly like the
+    # reference (Adam built without it, few_shot_learning_system.py:69;
+    # all shipped configs set 0.0) — see PARITY.md "conscious deviations"
     p.add_argument("--weight_decay", type=float, default=0.0)
 
     # --- misc reference flags ---

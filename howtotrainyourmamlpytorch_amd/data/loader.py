@@ -43,7 +43,11 @@ class MetaLearningSystemDataLoader:
         self.args = args
         self.rank = rank
         self.world_size = world_size
-        self.global_batch = args.batch_size
+        # samples_per_iter multiplies the tasks consumed per iteration,
+        # exactly like the reference's DataLoader batch_size multiplier
+        # (data.py:575-581); num_of_gpus is NOT multiplied in — GPU count
+        # is carried by world_size sharding instead.
+        self.global_batch = args.batch_size * int(getattr(args, "samples_per_iter", 1) or 1)
         if self.global_batch % world_size != 0:
             raise ValueError(f"batch_size {self.global_batch} not divisible by "
                              f"world_size {world_size}")

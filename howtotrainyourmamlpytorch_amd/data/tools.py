@@ -5,6 +5,7 @@ mini-imagenet 100 x 600)."""
 
 from __future__ import annotations
 
+import json
 import os
 import shutil
 import subprocess
@@ -51,3 +52,21 @@ def maybe_unzip_dataset(args) -> None:
     if os.path.isdir(path):
         shutil.rmtree(path)
     unzip_file(archive, os.path.dirname(os.path.abspath(path)) or ".")
+
+
+def export_label_maps(dataset_name: str, class_names, out_dir: str) -> tuple:
+    """Write the reference's label-map JSON pair next to a dataset
+    (``datasets/label_name_to_map_<ds>.json`` mapping class name -> index
+    and ``map_to_label_name_<ds>.json`` mapping index -> class name —
+    reference files ``/root/reference/datasets/label_name_to_map_*.json``).
+    Returns the two paths."""
+    os.makedirs(out_dir, exist_ok=True)
+    name_to_idx = {str(n): i for i, n in enumerate(class_names)}
+    idx_to_name = {str(i): str(n) for i, n in enumerate(class_names)}
+    p1 = os.path.join(out_dir, f"label_name_to_map_{dataset_name}.json")
+    p2 = os.path.join(out_dir, f"map_to_label_name_{dataset_name}.json")
+    with open(p1, "w") as f:
+        json.dump(name_to_idx, f)
+    with open(p2, "w") as f:
+        json.dump(idx_to_name, f)
+    return p1, p2

@@ -98,20 +98,35 @@ class ExperimentBuilder:
         return losses
 
     # ------------------------------------------------------------------
+    def _dist_mean_std(self, vals: List[float]) -> tuple:
+        """World-size-invariant mean/std: all ranks contribute their local
+        per-batch values via an all-reduced (count, sum, sum-of-squares)
+        triple, so the logged statistics describe the full evaluation set
+        regardless of GPU count."""
+        n = float(len(vals))
+        s = float(np.sum(vals)) if vals else 0.0
+        q = float(np.sum(np.square(vals))) if vals else 0.0
+        if self.dist is not None and self.world_size > 1:
+            n, s, q = self.dist.all_reduce_sum_vector([n, s, q])
+        if n == 0:
+            return 0.0, 0.0
+        mean = s / n
+        var = max(0.0, q / n - mean * mean)
+        return mean, float(np.sqrt(var))
+
     def _epoch_summary(self, val_losses: List[Dict[str, float]],
                        epoch: int, epoch_time: float) -> Dict[str, float]:
         summary: Dict[str, float] = {"epoch": epoch, "epoch_run_time": epoch_time}
         for key in ("train_loss", "train_accuracy"):
             vals = self.total_losses.get(key, [])
             if vals:
-                summary[f"{key}_mean"] = float(np.mean(vals))
-                summary[f"{key}_std"] = float(np.std(vals))
+                m, s = self._dist_mean_std(vals)
+                summary[f"{key}_mean"] = m
+                summary[f"{key}_std"] = s
         for key in ("loss", "accuracy"):
             vals = [l[key] for l in val_losses if key in l]
             if vals:
-                m, s = float(np.mean(vals)), float(np.std(vals))
-                if self.dist is not None and self.world_size > 1:
-                    m = self.dist.all_reduce_scalar(m)
+                m, s = self._dist_mean_std(vals)
                 summary[f"val_{key}_mean"] = m
                 summary[f"val_{key}_std"] = s
         self.total_losses = defaultdict(list)
@@ -125,13 +140,20 @@ class ExperimentBuilder:
                 os.path.join(self.saved_models_filepath, name), state=self.state)
 
     def pack_and_save_metrics(self, summary: Dict[str, float], first: bool) -> None:
-        if self.rank != 0:
-            return
+        # per_epoch_statistics is maintained on EVERY rank (the final
+        # ensemble test ranks epochs from it on all ranks and participates
+        # symmetrically in the collectives); only the file writes are
+        # rank-0-gated.
         stats = self.state["per_epoch_statistics"]
         for k, v in summary.items():
             stats[k].append(v)
+        if self.rank != 0:
+            return
         header = sorted(summary.keys())
-        if first:
+        csv_path = os.path.join(self.logs_filepath, "summary_statistics.csv")
+        if first or not os.path.isfile(csv_path):
+            # header whenever the file is missing, not only at epoch 0 —
+            # resuming into a fresh logs dir must not yield headerless rows
             save_statistics(self.logs_filepath, header, create=True)
         save_statistics(self.logs_filepath, [summary.get(k, "") for k in header])
         save_to_json(os.path.join(self.logs_filepath, "summary_statistics.json"),
@@ -204,8 +226,13 @@ class ExperimentBuilder:
         (reference: ``experiment_builder.py:247-300``): per model, softmax
         target predictions per task; average across models; argmax; report
         accuracy mean/std."""
+        if self.dist is not None and self.world_size > 1:
+            # rank 0 writes the checkpoints every rank is about to load
+            self.dist.barrier()
         stats = self.state["per_epoch_statistics"]
         if not stats.get("val_accuracy_mean"):
+            # consistent across ranks: per_epoch_statistics is maintained
+            # on every rank, so either all ranks skip or none do
             self._print("no per-epoch stats; skipping ensemble test")
             return None
         val_acc = np.array(stats["val_accuracy_mean"], dtype=np.float64)
@@ -233,10 +260,14 @@ class ExperimentBuilder:
         ensemble = torch.stack(per_model_preds, dim=0).mean(dim=0)
         pred_labels = ensemble.argmax(dim=-1)
         per_task_acc = (pred_labels == labels).float().mean(dim=1)
-        acc_mean = float(per_task_acc.mean())
-        acc_std = float(per_task_acc.std())
+        # world-size-invariant mean/std over ALL ranks' task shards
+        n = float(per_task_acc.numel())
+        s = float(per_task_acc.sum())
+        q = float((per_task_acc * per_task_acc).sum())
         if self.dist is not None and self.world_size > 1:
-            acc_mean = self.dist.all_reduce_scalar(acc_mean)
+            n, s, q = self.dist.all_reduce_sum_vector([n, s, q])
+        acc_mean = s / max(n, 1.0)
+        acc_std = float(np.sqrt(max(0.0, q / max(n, 1.0) - acc_mean * acc_mean)))
         result = {"test_accuracy_mean": acc_mean, "test_accuracy_std": acc_std}
         if self.rank == 0:
             save_statistics(self.logs_filepath,

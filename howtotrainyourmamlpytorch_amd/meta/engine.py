@@ -247,29 +247,40 @@ class MAMLFewShotClassifier(nn.Module):
         single unrolled graph fits in HBM."""
         total = data_batch[0].shape[0]
         self.optimizer.zero_grad(set_to_none=True)
-        agg_loss = 0.0
-        agg_acc = 0.0
+        agg: Dict[str, float] = {}
+        weights_sum = 0.0
         preds = []
-        base_losses = None
+        overlap = (self.dist is not None and self.dist.world_size > 1)
+        if overlap:
+            self.dist.start_overlapped_reduction(self.trainable_parameters())
         for lo in range(0, total, chunk):
             hi = min(lo + chunk, total)
             sub = tuple(x[lo:hi] for x in data_batch)
             losses, p = self.train_forward_prop(sub, epoch)
-            (losses["loss"] * ((hi - lo) / total)).backward()
-            agg_loss += float(losses["loss"].detach()) * (hi - lo) / total
-            agg_acc += losses["accuracy"] * (hi - lo) / total
+            wgt = (hi - lo) / total
+            (losses["loss"] * wgt).backward()
+            if overlap:
+                # north-star overlap: all-reduce this chunk's gradient
+                # contribution on the comm stream while the next chunk's
+                # inner loop computes (all-reduce is linear, so per-chunk
+                # reduction == reduction of the accumulated sum)
+                self.dist.reduce_chunk_gradients(self.trainable_parameters())
+            # weighted mean over chunks for every scalar entry (the
+            # importance-vector entries are identical across chunks, so the
+            # weighted mean reproduces them exactly)
+            for k, v in losses.items():
+                val = float(v.detach()) if torch.is_tensor(v) else float(v)
+                agg[k] = agg.get(k, 0.0) + val * wgt
+            weights_sum += wgt
             preds.append(p)
-            base_losses = losses
-        if self.dist is not None and self.dist.world_size > 1:
-            self.dist.all_reduce_gradients(self.trainable_parameters())
+        if overlap:
+            self.dist.finish_overlapped_reduction(self.trainable_parameters())
         if "imagenet" in self.args.dataset_name:
             for p_ in self.trainable_parameters():
                 if p_.grad is not None:
                     p_.grad.data.clamp_(-10, 10)
         self.optimizer.step()
-        base_losses["loss"] = agg_loss
-        base_losses["accuracy"] = agg_acc
-        return base_losses, torch.cat(preds, dim=0)
+        return agg, torch.cat(preds, dim=0)
 
     def run_train_iter(self, data_batch, epoch):
         epoch = int(epoch)

@@ -96,7 +96,10 @@ class TaskBatchedVGG(nn.Module):
             named_shapes.append((f"layer_dict.conv{i}.conv.weight", (num_filters, cin, 3, 3)))
             named_shapes.append((f"layer_dict.conv{i}.conv.bias", (num_filters,)))
             if inner_loop_bn_params and norm_layer == "batch_norm":
-                bn_shape = (num_steps, num_filters) if per_step_bn_statistics else (num_filters,)
+                # reference overrides per-step gamma/beta to a single
+                # [num_features] parameter when BN params are inner-loop
+                # fast weights (meta_neural_network_architectures.py:194-198)
+                bn_shape = (num_filters,)
                 named_shapes.append((f"layer_dict.conv{i}.norm_layer.weight", bn_shape))
                 named_shapes.append((f"layer_dict.conv{i}.norm_layer.bias", bn_shape))
         # linear head: weight stored [ways, K] with K = NHWC-flattened feature
@@ -111,7 +114,7 @@ class TaskBatchedVGG(nn.Module):
             init[f"layer_dict.conv{i}.conv.weight"] = _xavier_uniform((num_filters, cin, 3, 3), generator)
             init[f"layer_dict.conv{i}.conv.bias"] = torch.zeros(num_filters)
             if inner_loop_bn_params and norm_layer == "batch_norm":
-                bn_shape = (num_steps, num_filters) if per_step_bn_statistics else (num_filters,)
+                bn_shape = (num_filters,)
                 init[f"layer_dict.conv{i}.norm_layer.weight"] = torch.ones(*bn_shape)
                 init[f"layer_dict.conv{i}.norm_layer.bias"] = torch.zeros(*bn_shape)
         init["layer_dict.linear.weights"] = _xavier_uniform(
@@ -225,12 +228,9 @@ class TaskBatchedVGG(nn.Module):
         weights are used."""
         name_w = f"layer_dict.conv{i}.norm_layer.weight"
         if self.inner_loop_bn_params and name_w in v:
-            gamma = v[name_w]
-            beta = v[f"layer_dict.conv{i}.norm_layer.bias"]
-            if self.per_step_bn_statistics:
-                gamma = gamma[:, num_step]   # [T, F]
-                beta = beta[:, num_step]
-            return gamma, beta
+            # single [T, F] fast weight regardless of per-step stats
+            # (reference override, meta_neural_network_architectures.py:194-198)
+            return v[name_w], v[f"layer_dict.conv{i}.norm_layer.bias"]
         gamma = getattr(self, f"bn_weight_{i}")
         beta = getattr(self, f"bn_bias_{i}")
         if self.per_step_bn_statistics:

@@ -358,10 +358,29 @@ class _ConvWgradFn(torch.autograd.Function):
         return d_dy, d_x, None, None
 
 
+_FALLBACK_WARNED = set()
+
+
+def _warn_fallback(reason: str) -> None:
+    if reason not in _FALLBACK_WARNED:
+        _FALLBACK_WARNED.add(reason)
+        import warnings
+        warnings.warn(
+            f"maml355: task_conv3x3 using the grouped-ATen composition on GPU "
+            f"({reason}); the MFMA kernel path does not cover this shape/config.",
+            stacklevel=3)
+
+
 def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
-    if stride != 1 or x.dtype != torch.bfloat16 or w.shape[1] > 64:
-        # stride-2 (max_pooling=False configs) and fp32 compute use the
-        # grouped-ATen composition; cast for dtype consistency
+    if stride != 1 or x.dtype != torch.bfloat16 or w.shape[0] > 64 or w.shape[1] > 64:
+        # stride-2 (max_pooling=False configs), fp32 compute and >64-channel
+        # shapes use the grouped-ATen composition; loud, not silent
+        if stride != 1:
+            _warn_fallback("stride != 1")
+        elif x.dtype != torch.bfloat16:
+            _warn_fallback(f"dtype {x.dtype}")
+        else:
+            _warn_fallback(f"channels > 64 (F={w.shape[0]}, C={w.shape[1]})")
         wc = w.to(x.dtype)
         bc = b.to(x.dtype) if b is not None else None
         y = ref.task_conv3x3(x, wc, bc, stride, padding)

@@ -60,6 +60,67 @@ class DistContext:
             g.copy_(flat[off:off + g.numel()].view_as(g))
             off += g.numel()
 
+    # ------------------------------------------------------------------
+    # Overlapped per-chunk reduction (the BASELINE.json north-star overlap):
+    # with task-chunked gradient accumulation, all-reduce is linear in the
+    # chunks, so each chunk's gradient contribution is all-reduced
+    # asynchronously (RCCL runs it on its own comm stream) while the next
+    # chunk's inner loop computes on the main stream.  finish() waits,
+    # sums the reduced buckets and writes the averaged result into .grad.
+    def start_overlapped_reduction(self, params: List[torch.Tensor]) -> None:
+        self._ov_pending = []   # list of (work_handle, flat_buffer)
+
+    def reduce_chunk_gradients(self, params: List[torch.Tensor]) -> None:
+        total = sum(p.numel() for p in params)
+        device = params[0].device
+        pool = getattr(self, "_ov_pool", [])
+        buf = pool.pop() if pool else torch.empty(
+            total, dtype=torch.float32, device=device)
+        self._ov_pool = pool
+        off = 0
+        for p in params:
+            n = p.numel()
+            if p.grad is None:
+                buf[off:off + n].zero_()
+            else:
+                buf[off:off + n].copy_(p.grad.reshape(-1))
+                p.grad.zero_()   # next chunk accumulates from zero
+            off += n
+        work = dist.all_reduce(buf, op=dist.ReduceOp.SUM, async_op=True)
+        self._ov_pending.append((work, buf))
+
+    def finish_overlapped_reduction(self, params: List[torch.Tensor]) -> None:
+        total_buf = None
+        for work, buf in self._ov_pending:
+            work.wait()
+            if total_buf is None:
+                total_buf = buf
+            else:
+                total_buf.add_(buf)
+                self._ov_pool.append(buf)
+        self._ov_pending = []
+        if total_buf is None:
+            return
+        total_buf.div_(self.world_size)
+        off = 0
+        for p in params:
+            n = p.numel()
+            if p.grad is None:
+                p.grad = torch.zeros_like(p)
+            p.grad.copy_(total_buf[off:off + n].view_as(p))
+            off += n
+        self._ov_pool.append(total_buf)
+
+    def all_reduce_sum_vector(self, values) -> list:
+        """SUM-all-reduce a small list of python floats in one collective;
+        returns the reduced list.  Used for world-size-invariant
+        (count, sum, sum-of-squares) statistics triples."""
+        t = torch.tensor(list(values), dtype=torch.float64)
+        if self.backend == "nccl":
+            t = t.to(torch.device("cuda", self.local_rank))
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        return [float(v) for v in t.tolist()]
+
     def all_reduce_scalar(self, value: float, average: bool = True) -> float:
         t = torch.tensor([value], dtype=torch.float64)
         if self.backend == "nccl":

@@ -29,6 +29,12 @@ def main(root, out):
             images[ci, si] = np.asarray(img, dtype=np.uint8)
     np.savez_compressed(out, images=images,
                         class_names=np.array(names, dtype=object))
+    # label-map JSON pair beside the npz (reference ships
+    # datasets/label_name_to_map_*.json / map_to_label_name_*.json)
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    from howtotrainyourmamlpytorch_amd.data.tools import export_label_maps
+    ds_name = os.path.splitext(os.path.basename(out))[0]
+    export_label_maps(ds_name, names, os.path.dirname(os.path.abspath(out)))
     print(f"{len(names)} classes x {spc} samples -> {out} "
           f"({os.path.getsize(out)/1e6:.1f} MB)")
 
