@@ -76,8 +76,12 @@ class MAMLFewShotClassifier(nn.Module):
             requires_grad=learnable)
 
         self.to(device)
-        self.optimizer = torch.optim.Adam(self.trainable_parameters(),
-                                          lr=args.meta_learning_rate, amsgrad=False)
+        # fused multi-tensor Adam: one launch per meta-update, with the
+        # reference's imagenet ±10 grad clamp fused in
+        from .fused_adam import FusedAdam
+        clamp = 10.0 if "imagenet" in args.dataset_name else None
+        self.optimizer = FusedAdam(self.trainable_parameters(),
+                                   lr=args.meta_learning_rate, grad_clamp=clamp)
         self.dist = None  # set by attach_distributed()
         self.timers = None
         if getattr(args, "enable_phase_timers", False):
@@ -230,10 +234,7 @@ class MAMLFewShotClassifier(nn.Module):
         loss.backward()
         if self.dist is not None and self.dist.world_size > 1:
             self.dist.all_reduce_gradients(self.trainable_parameters())
-        if "imagenet" in self.args.dataset_name:
-            for p in self.trainable_parameters():
-                if p.grad is not None:
-                    p.grad.data.clamp_(-10, 10)
+        # imagenet grad clamp is fused into the Adam kernel (FusedAdam)
         self.optimizer.step()
 
     def _chunked_train_step(self, data_batch, epoch: int, chunk: int):
@@ -275,10 +276,6 @@ class MAMLFewShotClassifier(nn.Module):
             preds.append(p)
         if overlap:
             self.dist.finish_overlapped_reduction(self.trainable_parameters())
-        if "imagenet" in self.args.dataset_name:
-            for p_ in self.trainable_parameters():
-                if p_.grad is not None:
-                    p_.grad.data.clamp_(-10, 10)
         self.optimizer.step()
         return agg, torch.cat(preds, dim=0)
 

@@ -44,6 +44,13 @@ std::vector<torch::Tensor> tconv_mm(torch::Tensor x, torch::Tensor wp,
 std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
                                        long pad, bool with_bias);
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B);
+// adam.hip
+void adam_step(std::vector<torch::Tensor> params,
+               std::vector<torch::Tensor> grads,
+               std::vector<torch::Tensor> exp_avgs,
+               std::vector<torch::Tensor> exp_avg_sqs,
+               long step, double lr, double beta1, double beta2, double eps,
+               double weight_decay, double clamp_v);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd, "fused task-batched BN+leakyReLU fwd");
@@ -62,4 +69,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tconv_mm", &tconv_mm, "task-batched MFMA 3x3 conv fwd/dgrad");
   m.def("tconv_wgrad", &tconv_wgrad, "task-batched MFMA 3x3 conv wgrad");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("adam_step", &adam_step, "fused multi-tensor Adam + grad clamp");
 }

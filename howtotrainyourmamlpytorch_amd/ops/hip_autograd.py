@@ -397,5 +397,15 @@ def task_linear(x, w, b=None):
 
 def fused_adam_step(params, grads, exp_avgs, exp_avg_sqs, step, lr,
                     beta1=0.9, beta2=0.999, eps=1e-8, weight_decay=0.0, clamp=None):
-    return ref.fused_adam_step(params, grads, exp_avgs, exp_avg_sqs, step, lr,
-                               beta1, beta2, eps, weight_decay, clamp)
+    """One launch for the whole meta-update: multi-tensor Adam with the
+    reference's pre-clamp fused in (few_shot_learning_system.py:330-336)."""
+    keep = [i for i, g in enumerate(grads) if g is not None]
+    if not keep:
+        return
+    ps = [params[i].data.view(-1) for i in keep]
+    gs = [grads[i].contiguous().view(-1).float() for i in keep]
+    ms = [exp_avgs[i].view(-1) for i in keep]
+    vs = [exp_avg_sqs[i].view(-1) for i in keep]
+    _ext().adam_step(ps, gs, ms, vs, int(step), float(lr), float(beta1),
+                     float(beta2), float(eps), float(weight_decay),
+                     float(clamp) if clamp is not None else -1.0)
