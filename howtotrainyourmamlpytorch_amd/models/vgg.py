@@ -177,9 +177,12 @@ class TaskBatchedVGG(nn.Module):
             # conv-epilogue BN-stats fusion measured NET-NEGATIVE (-9%:
             # LDS atomics in the hot conv kernel cost more than the cheap
             # separate stats pass) — keep the capability but default off
+            # (epilogue fusion uses LDS/global atomics, so it is disabled
+            # under the deterministic-reduction mode)
             want_stats = (self.norm_layer_type == "batch_norm"
                           and self.max_pooling
-                          and os.environ.get("MAML355_EPIFUSE", "0") == "1")
+                          and os.environ.get("MAML355_EPIFUSE", "0") == "1"
+                          and os.environ.get("MAML355_DETERMINISTIC", "0") != "1")
             bn_sums = None
             if want_stats:
                 out, bn_sums = ops.task_conv3x3(out, w_i, b_i, stride=stride,

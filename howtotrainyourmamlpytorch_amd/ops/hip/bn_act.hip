@@ -13,9 +13,14 @@
 
 using namespace maml355;
 
+// Partial sums land in partials[T, NBLK, 2, C] (one private slice per
+// block — NO global atomics) and are reduced over NBLK in fixed order by
+// bn_reduce2_kernel / bn_finalize_partials_kernel.  This makes every BN
+// reduction bitwise run-to-run deterministic and removes
+// the global-atomic contention of the round-1 kernels.
 template <typename scalar_t>
 __global__ void bn_sums_vec_kernel(const scalar_t* __restrict__ x,
-                                   float* __restrict__ sums,  // [T, 2, C]
+                                   float* __restrict__ partials,  // [T,NBLK,2,C]
                                    int T, long M, int C, int rows_per_block) {
   const int c8n = C / 8;
   const int rows_in_block = blockDim.x / c8n;
@@ -23,11 +28,9 @@ __global__ void bn_sums_vec_kernel(const scalar_t* __restrict__ x,
   const int rg = threadIdx.x / c8n;
   const int t = blockIdx.x;
   const long row0 = (long)blockIdx.y * rows_per_block;
-  extern __shared__ float ls[];  // [2][C]
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
-  __syncthreads();
+  extern __shared__ float ls[];  // [rows_in_block][2][C] per-thread slices
+  float s[8] = {0}, q[8] = {0};
   if (rg < rows_in_block) {
-    float s[8] = {0}, q[8] = {0};
     const scalar_t* xt = x + (long)t * M * C + c8 * 8;
     const long row_end = min(row0 + rows_per_block, M);
     for (long m = row0 + rg; m < row_end; m += rows_in_block) {
@@ -38,15 +41,46 @@ __global__ void bn_sums_vec_kernel(const scalar_t* __restrict__ x,
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&ls[c8 * 8 + j], s[j]);
-      atomicAdd(&ls[C + c8 * 8 + j], q[j]);
+      ls[(rg * 2 + 0) * C + c8 * 8 + j] = s[j];
+      ls[(rg * 2 + 1) * C + c8 * 8 + j] = q[j];
     }
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    atomicAdd(&sums[((long)t * 2 + 0) * C + i], ls[i]);
-    atomicAdd(&sums[((long)t * 2 + 1) * C + i], ls[C + i]);
+  // in-block reduce in FIXED row-group order (bitwise deterministic)
+  if (rg == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { s[j] = 0.f; q[j] = 0.f; }
+    for (int rr = 0; rr < rows_in_block; ++rr) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s[j] += ls[(rr * 2 + 0) * C + c8 * 8 + j];
+        q[j] += ls[(rr * 2 + 1) * C + c8 * 8 + j];
+      }
+    }
+    float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 2) * C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pt[c8 * 8 + j] = s[j];
+      pt[C + c8 * 8 + j] = q[j];
+    }
   }
+}
+
+// ordered reduction of partials[T, NBLK, 2, C] -> sums[T, 2, C]
+__global__ void bn_reduce2_kernel(const float* __restrict__ partials,
+                                  float* __restrict__ sums, int T, int nblk,
+                                  int C) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= T * C) return;
+  const int t = i / C, c = i % C;
+  float s = 0.f, q = 0.f;
+  for (int b = 0; b < nblk; ++b) {
+    const float* p = partials + (((long)t * nblk + b) * 2) * C;
+    s += p[c];
+    q += p[C + c];
+  }
+  sums[((long)t * 2 + 0) * C + c] = s;
+  sums[((long)t * 2 + 1) * C + c] = q;
 }
 
 template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
@@ -152,6 +186,7 @@ __global__ void bn_norm_act_pool_vec_kernel(
   }
 }
 
+// deterministic in-block + cross-block reduction (see bn_sums_vec_kernel)
 template <typename scalar_t, bool PER_TASK_AFFINE, bool ACT>
 __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
                                        const scalar_t* __restrict__ x,
@@ -159,7 +194,7 @@ __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
                                        const float* __restrict__ rstd,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ beta,
-                                       float* __restrict__ bsums,  // [T, 2, C]
+                                       float* __restrict__ partials,  // [T,NBLK,2,C]
                                        int T, long M, int C, float slope,
                                        int rows_per_block) {
   const int c8n = C / 8;
@@ -168,9 +203,8 @@ __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
   const int rg = threadIdx.x / c8n;
   const int t = blockIdx.x;
   const long row0 = (long)blockIdx.y * rows_per_block;
-  extern __shared__ float ls[];  // [2][C]
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
-  __syncthreads();
+  extern __shared__ float ls[];  // [rows_in_block][2][C]
+  float s1[8] = {0}, s2[8] = {0};
   if (rg < rows_in_block) {
     float mu[8], r[8], g[8], b[8];
 #pragma unroll
@@ -182,7 +216,6 @@ __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
       g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c];
       b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c];
     }
-    float s1[8] = {0}, s2[8] = {0};
     const scalar_t* xt = x + (long)t * M * C + c8 * 8;
     const scalar_t* dyt = dy + (long)t * M * C + c8 * 8;
     const long row_end = min(row0 + rows_per_block, M);
@@ -201,14 +234,27 @@ __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&ls[c8 * 8 + j], s1[j]);
-      atomicAdd(&ls[C + c8 * 8 + j], s2[j]);
+      ls[(rg * 2 + 0) * C + c8 * 8 + j] = s1[j];
+      ls[(rg * 2 + 1) * C + c8 * 8 + j] = s2[j];
     }
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    atomicAdd(&bsums[((long)t * 2 + 0) * C + i], ls[i]);
-    atomicAdd(&bsums[((long)t * 2 + 1) * C + i], ls[C + i]);
+  if (rg == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { s1[j] = 0.f; s2[j] = 0.f; }
+    for (int rr = 0; rr < rows_in_block; ++rr) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s1[j] += ls[(rr * 2 + 0) * C + c8 * 8 + j];
+        s2[j] += ls[(rr * 2 + 1) * C + c8 * 8 + j];
+      }
+    }
+    float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 2) * C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pt[c8 * 8 + j] = s1[j];
+      pt[C + c8 * 8 + j] = s2[j];
+    }
   }
 }
 
@@ -393,7 +439,7 @@ __global__ void bn_pool_bwd_sums_vec_kernel(
     const scalar_t* __restrict__ x,        // [T, NB, H, W, C]
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
-    float* __restrict__ bsums,             // [T, 2, C]
+    float* __restrict__ partials,          // [T, NBLK, 2, C]
     int T, int NB, int H, int W, int C, float slope, int rows_per_block) {
   const long M = (long)NB * H * W;
   const int c8n = C / 8;
@@ -402,11 +448,10 @@ __global__ void bn_pool_bwd_sums_vec_kernel(
   const int rg = threadIdx.x / c8n;
   const int t = blockIdx.x;
   const long row0 = (long)blockIdx.y * rows_per_block;
-  extern __shared__ float ls[];
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
-  __syncthreads();
+  extern __shared__ float ls[];  // [rows_in_block][2][C]
+  float s1[8] = {0}, s2[8] = {0};
+  const int c0 = c8 * 8;
   if (rg < rows_in_block) {
-    const int c0 = c8 * 8;
     float mu[8], r[8], g[8], b[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -415,7 +460,6 @@ __global__ void bn_pool_bwd_sums_vec_kernel(
       g[j] = PER_TASK_AFFINE ? gamma[tc] : gamma[c0 + j];
       b[j] = PER_TASK_AFFINE ? beta[tc] : beta[c0 + j];
     }
-    float s1[8] = {0}, s2[8] = {0};
     const scalar_t* xt = x + (long)t * M * C + c0;
     const scalar_t* dypt = dyp + (long)t * NB * (H / 2) * (W / 2) * C;
     const unsigned char* mt_ = mask + (long)t * NB * (H / 2) * (W / 2) * C;
@@ -435,14 +479,27 @@ __global__ void bn_pool_bwd_sums_vec_kernel(
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&ls[c0 + j], s1[j]);
-      atomicAdd(&ls[C + c0 + j], s2[j]);
+      ls[(rg * 2 + 0) * C + c0 + j] = s1[j];
+      ls[(rg * 2 + 1) * C + c0 + j] = s2[j];
     }
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    atomicAdd(&bsums[((long)t * 2 + 0) * C + i], ls[i]);
-    atomicAdd(&bsums[((long)t * 2 + 1) * C + i], ls[C + i]);
+  if (rg == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { s1[j] = 0.f; s2[j] = 0.f; }
+    for (int rr = 0; rr < rows_in_block; ++rr) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s1[j] += ls[(rr * 2 + 0) * C + c0 + j];
+        s2[j] += ls[(rr * 2 + 1) * C + c0 + j];
+      }
+    }
+    float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 2) * C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pt[c0 + j] = s1[j];
+      pt[C + c0 + j] = s2[j];
+    }
   }
 }
 
@@ -617,12 +674,18 @@ void bn_fwd_impl(const torch::Tensor& x, const torch::Tensor& gamma,
   const int threads = cpad * std::max<int>(1, kThreads / cpad);
   const int lds_bytes = 2 * threads * sizeof(float);
   if (vec) {
-    const int rpb = 1024;  // amortize the per-block LDS/global atomic rounds
-    dim3 g(T, (unsigned)((M + rpb - 1) / rpb));
+    const int rpb = 1024;  // amortize the per-block staging rounds
+    const int nb = (int)((M + rpb - 1) / rpb);
+    dim3 g(T, (unsigned)nb);
+    auto partials = torch::empty({T, nb, 2, C},
+                                 x.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL((bn_sums_vec_kernel<scalar_t>), g, dim3(256),
-                       2 * C * (int)sizeof(float), stream.stream(),
+                       16384, stream.stream(),
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
-                       sums.data_ptr<float>(), T, M, C, rpb);
+                       partials.data_ptr<float>(), T, M, C, rpb);
+    hipLaunchKernelGGL(bn_reduce2_kernel, dim3((T * C + 255) / 256), dim3(256),
+                       0, stream.stream(), partials.data_ptr<float>(),
+                       sums.data_ptr<float>(), T, nb, C);
   } else
   hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), sums_grid, dim3(threads),
                      lds_bytes, stream.stream(),
@@ -678,16 +741,23 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& x,
 #define LAUNCH_BWD(PT, ACT_)                                                   \
   do {                                                                         \
     if (vec) {                                                                 \
-      dim3 gs(T, (unsigned)((M + 1023) / 1024));                               \
+      const int nb = (int)((M + 1023) / 1024);                                 \
+      dim3 gs(T, (unsigned)nb);                                                \
+      auto partials = torch::empty({T, nb, 2, C},                              \
+                                   x.options().dtype(torch::kFloat32));        \
       hipLaunchKernelGGL((bn_bwd_sums_vec_kernel<scalar_t, PT, ACT_>),         \
-                         gs, dim3(256), 2 * C * (int)sizeof(float),            \
+                         gs, dim3(256), 16384,                                 \
                          stream.stream(),                                      \
                          reinterpret_cast<const scalar_t*>(dy.data_ptr()),     \
                          reinterpret_cast<const scalar_t*>(x.data_ptr()),      \
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
                          gamma.data_ptr<float>(), beta.data_ptr<float>(),      \
-                         bsums.data_ptr<float>(), T, M, C, (float)slope,       \
+                         partials.data_ptr<float>(), T, M, C, (float)slope,    \
                          1024);                                                \
+      hipLaunchKernelGGL(bn_reduce2_kernel, dim3((T * C + 255) / 256),         \
+                         dim3(256), 0, stream.stream(),                        \
+                         partials.data_ptr<float>(),                           \
+                         bsums.data_ptr<float>(), T, nb, C);                   \
       hipLaunchKernelGGL((bn_bwd_dx_vec_kernel<scalar_t, PT, ACT_>),           \
                          sums_grid, dim3(256), 0, stream.stream(),             \
                          reinterpret_cast<const scalar_t*>(dy.data_ptr()),     \
@@ -812,10 +882,16 @@ std::vector<torch::Tensor> bn_act_pool_fwd(torch::Tensor x, torch::Tensor gamma,
 #define LAUNCH_BNP(ST, PT)                                                     \
   do {                                                                         \
     if (!have_sums) {                                                          \
+      const int nb_ = (int)((M + rpb - 1) / rpb);                              \
+      auto partials = torch::empty({T, nb_, 2, C}, fopts);                     \
       hipLaunchKernelGGL((bn_sums_vec_kernel<ST>), gsums, dim3(256),           \
-                         2 * C * (int)sizeof(float), stream.stream(),          \
+                         16384, stream.stream(),                               \
                          reinterpret_cast<const ST*>(x.data_ptr()),            \
-                         sums.data_ptr<float>(), T, M, C, rpb);                \
+                         partials.data_ptr<float>(), T, M, C, rpb);            \
+      hipLaunchKernelGGL(bn_reduce2_kernel, dim3((T * C + 255) / 256),         \
+                         dim3(256), 0, stream.stream(),                        \
+                         partials.data_ptr<float>(),                           \
+                         sums.data_ptr<float>(), T, nb_, C);                   \
     }                                                                          \
     const int fin_blocks = (T * C + 255) / 256;                                \
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(fin_blocks), dim3(256), 0,     \
@@ -868,15 +944,21 @@ std::vector<torch::Tensor> bn_act_pool_bwd(torch::Tensor dyp, torch::Tensor mask
 
 #define LAUNCH_BPB(ST, PT)                                                     \
   do {                                                                         \
+    const int nb_ = (int)((M + rpb - 1) / rpb);                                \
+    auto partials = torch::empty({T, nb_, 2, C}, fopts);                       \
     hipLaunchKernelGGL((bn_pool_bwd_sums_vec_kernel<ST, PT>), g, dim3(256),    \
-                       2 * C * (int)sizeof(float), stream.stream(),            \
+                       16384, stream.stream(),                                 \
                        reinterpret_cast<const ST*>(dypc.data_ptr()),           \
                        mask.data_ptr<unsigned char>(),                         \
                        reinterpret_cast<const ST*>(x.data_ptr()),              \
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
                        gc.data_ptr<float>(), bc.data_ptr<float>(),             \
-                       bsums.data_ptr<float>(), T, NB, H, W, C, (float)slope,  \
-                       rpb);                                                   \
+                       partials.data_ptr<float>(), T, NB, H, W, C,             \
+                       (float)slope, rpb);                                     \
+    hipLaunchKernelGGL(bn_reduce2_kernel, dim3((T * C + 255) / 256),           \
+                       dim3(256), 0, stream.stream(),                          \
+                       partials.data_ptr<float>(),                             \
+                       bsums.data_ptr<float>(), T, nb_, C);                    \
     hipLaunchKernelGGL((bn_pool_bwd_dx_vec_kernel<ST, PT>), g, dim3(256), 0,   \
                        stream.stream(),                                        \
                        reinterpret_cast<const ST*>(dypc.data_ptr()),           \

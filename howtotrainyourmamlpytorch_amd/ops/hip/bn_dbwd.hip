@@ -39,7 +39,7 @@ __global__ void bn_dbwd_sums_kernel(const scalar_t* __restrict__ x,
                                     const float* __restrict__ rstd,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ beta,
-                                    float* __restrict__ sums,  // [T,5,C]
+                                    float* __restrict__ partials,  // [T,NBLK,5,C]
                                     int T, long M, int C, float slope,
                                     bool act, int rows_per_block) {
   const int cpad = ((C + WAVE - 1) / WAVE) * WAVE;
@@ -85,12 +85,32 @@ __global__ void bn_dbwd_sums_kernel(const scalar_t* __restrict__ x,
       Gxh += lds[3 * n + rr * cpad + c];
       Gu += lds[4 * n + rr * cpad + c];
     }
-    atomicAdd(&sums[((long)t * 5 + 0) * C + c], s1);
-    atomicAdd(&sums[((long)t * 5 + 1) * C + c], s2);
-    atomicAdd(&sums[((long)t * 5 + 2) * C + c], G);
-    atomicAdd(&sums[((long)t * 5 + 3) * C + c], Gxh);
-    atomicAdd(&sums[((long)t * 5 + 4) * C + c], Gu);
+    // private per-block slice; reduced in fixed order by bn_reduce5_kernel
+    float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 5) * C;
+    pt[0 * C + c] = s1;
+    pt[1 * C + c] = s2;
+    pt[2 * C + c] = G;
+    pt[3 * C + c] = Gxh;
+    pt[4 * C + c] = Gu;
   }
+}
+
+// ordered reduction of partials[T, NBLK, 5, C] -> sums[T, 5, C]
+// (bitwise run-to-run deterministic; no global atomics)
+__global__ void bn_reduce5_kernel(const float* __restrict__ partials,
+                                  float* __restrict__ sums, int T, int nblk,
+                                  int C) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= T * C) return;
+  const int t = i / C, c = i % C;
+  float acc[5] = {0.f, 0.f, 0.f, 0.f, 0.f};
+  for (int b = 0; b < nblk; ++b) {
+    const float* p = partials + (((long)t * nblk + b) * 5) * C;
+#pragma unroll
+    for (int k = 0; k < 5; ++k) acc[k] += p[k * C + c];
+  }
+#pragma unroll
+  for (int k = 0; k < 5; ++k) sums[((long)t * 5 + k) * C + c] = acc[k];
 }
 
 // ---------------------------------------------------------------------------
@@ -191,6 +211,8 @@ std::vector<torch::Tensor> bn_act_dbwd(torch::Tensor x, torch::Tensor u,
 
 #define LAUNCH_DB(ST, PT)                                                      \
   do {                                                                         \
+    const int nblk = (int)sums_grid.y;                                         \
+    auto partials = torch::empty({T, nblk, 5, C}, fopts);                      \
     hipLaunchKernelGGL((bn_dbwd_sums_kernel<ST, PT>), sums_grid,               \
                        dim3(threads), lds_bytes, stream.stream(),              \
                        reinterpret_cast<const ST*>(x.data_ptr()),              \
@@ -198,8 +220,12 @@ std::vector<torch::Tensor> bn_act_dbwd(torch::Tensor x, torch::Tensor u,
                        reinterpret_cast<const ST*>(gxc.data_ptr()),            \
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
                        gc.data_ptr<float>(), bc.data_ptr<float>(),             \
-                       sums.data_ptr<float>(), T, M, C, (float)slope, act,     \
+                       partials.data_ptr<float>(), T, M, C, (float)slope, act, \
                        kRowsPerBlockD);                                        \
+    hipLaunchKernelGGL(bn_reduce5_kernel, dim3((T * C + 255) / 256),           \
+                       dim3(256), 0, stream.stream(),                          \
+                       partials.data_ptr<float>(), sums.data_ptr<float>(),     \
+                       T, nblk, C);                                            \
     hipLaunchKernelGGL((bn_dbwd_apply_kernel<ST, PT>), dim3(grid_ew(total)),   \
                        dim3(kThreadsD), 0, stream.stream(),                    \
                        reinterpret_cast<const ST*>(x.data_ptr()),              \

@@ -75,11 +75,15 @@ __global__ void repack_weight_kernel(const float* __restrict__ w,
   }
 }
 
-// fp32 wgrad accumulator [T, 9C, F] -> dW [T, F, C, 3, 3] fp32
+// fp32 wgrad accumulator [T, nslices, 9C, F] -> dW [T, F, C, 3, 3] fp32;
+// K-chunk slices are summed in FIXED order (nslices=1 for the atomic
+// fast path; >1 under MAML355_DETERMINISTIC where each K-chunk block
+// owns a private slice so the reduction is bitwise reproducible).
 __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
                                       float* __restrict__ dw,
-                                      int T, int F, int C) {
+                                      int T, int F, int C, int nslices) {
   const long total = (long)T * F * C * 9;
+  const long ssz = (long)9 * C * F;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += grid_stride()) {
     long r = i;
@@ -88,7 +92,12 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
     const int c = (int)(r % C); r /= C;
     const int f = (int)(r % F); r /= F;
     const long t = r;
-    dw[i] = acc[((t * 9 + (long)(ky * 3 + kx)) * C + c) * F + f];
+    const long e = ((long)(ky * 3 + kx) * C + c) * F + f;
+    float v = 0.f;
+    for (int s = 0; s < nslices; ++s) {
+      v += acc[(t * nslices + s) * ssz + e];
+    }
+    dw[i] = v;
   }
 }
 
@@ -302,10 +311,14 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 
 #define WGN 64   // wgrad output columns per block (4 waves x 16)
 
+// DET=true: each (t, K-chunk) block writes a PRIVATE accumulator slice
+// (plain stores, no atomics) — deterministic; DET=false: fp32 atomicAdd
+// into the shared per-task accumulator (fast path).
+template <bool DET>
 __global__ __launch_bounds__(256, 2)
 void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
-                        float* __restrict__ dWacc,
-                        float* __restrict__ dBacc,  // [T, F] or nullptr
+                        float* __restrict__ dWacc,  // [T, nsl, 9C, F]
+                        float* __restrict__ dBacc,  // [T, nsl, F] or nullptr
                         int T, int NB, int H, int W, int C,
                         int Ho, int Wo, int F, int pad, int kchunk) {
   const int t = blockIdx.z;
@@ -449,14 +462,21 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     db_lds[db_q][db_f] = db_acc;
     __syncthreads();
     if (db_q == 0 && db_f < F) {
-      atomicAdd(&dBacc[(long)t * F + db_f],
-                db_lds[0][db_f] + db_lds[1][db_f] + db_lds[2][db_f] +
-                    db_lds[3][db_f]);
+      const float v = db_lds[0][db_f] + db_lds[1][db_f] + db_lds[2][db_f] +
+                      db_lds[3][db_f];
+      if (DET) {
+        dBacc[((long)t * gridDim.y + blockIdx.y) * F + db_f] = v;
+      } else {
+        atomicAdd(&dBacc[(long)t * F + db_f], v);
+      }
     }
   }
 
   // accumulate: C/D col = lane&15 (n within wave tile), row = fk*4+j (f)
-  float* dWt = dWacc + (long)t * N9 * F;
+  // (n-columns are disjoint across blockIdx.x, so under DET the private
+  // (t, K-chunk) slice needs no atomics at all)
+  float* dWt = DET ? dWacc + ((long)t * gridDim.y + blockIdx.y) * N9 * F
+                   : dWacc + (long)t * N9 * F;
 #pragma unroll
   for (int mt = 0; mt < 4; ++mt) {
     if (mt >= mtiles) break;
@@ -465,7 +485,11 @@ void tconv_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
       const int f = mt * 16 + fk * 4 + j;
       const int n = n0 + wave * 16 + fr;
       if (f < F && n < N9) {
-        atomicAdd(&dWt[(long)n * F + f], acc[mt][j]);
+        if (DET) {
+          dWt[(long)n * F + f] = acc[mt][j];
+        } else {
+          atomicAdd(&dWt[(long)n * F + f], acc[mt][j]);
+        }
       }
     }
   }
@@ -567,29 +591,44 @@ std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
   const int Ho = (int)dy.size(2), Wo = (int)dy.size(3), F = (int)dy.size(4);
   TORCH_CHECK(F <= 64, "F must be <= 64");
   const int N9 = 9 * C;
-  auto acc = torch::zeros({T, N9, F}, x.options().dtype(torch::kFloat32));
-  auto db = torch::zeros({T, F}, x.options().dtype(torch::kFloat32));
   const long Ktot = (long)NB * Ho * Wo;
+  const char* det_env = getenv("MAML355_DETERMINISTIC");
+  const bool det = det_env && det_env[0] == '1';
   // size K-chunks so the grid has >= ~1024 blocks (256 CUs want far more
   // workgroups than CUs; small support-pass K was leaving the chip idle)
   const int gridx = (N9 + WGN - 1) / WGN;
   long desired_y = std::max<long>(1, 1024 / std::max<long>(1, (long)gridx * T));
   long kchunk = (Ktot + desired_y - 1) / desired_y;
   kchunk = ((kchunk + WBK - 1) / WBK) * WBK;
-  kchunk = std::min<long>(std::max<long>(kchunk, WBK), WG_KCHUNK);
-  dim3 grid((unsigned)gridx, (unsigned)((Ktot + kchunk - 1) / kchunk), T);
+  kchunk = std::max<long>(kchunk, WBK);
+  if (!det) kchunk = std::min<long>(kchunk, WG_KCHUNK);
+  const int gridy = (int)((Ktot + kchunk - 1) / kchunk);
+  // MAML355_DETERMINISTIC: each (t, K-chunk) block writes a private
+  // accumulator slice (no fp32 atomics); slices are summed in fixed order
+  // by the finalize kernel -> bitwise run-to-run reproducible wgrad.
+  const int nslices = det ? gridy : 1;
+  auto acc = torch::zeros({T, nslices, N9, F},
+                          x.options().dtype(torch::kFloat32));
+  auto dbacc = torch::zeros({T, nslices, F},
+                            x.options().dtype(torch::kFloat32));
+  dim3 grid((unsigned)gridx, (unsigned)gridy, T);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(tconv_wgrad_kernel, grid, dim3(256), 0, stream.stream(),
-                     reinterpret_cast<const bf16*>(dyc.data_ptr()),
-                     reinterpret_cast<const bf16*>(xc.data_ptr()),
-                     acc.data_ptr<float>(),
-                     with_bias ? db.data_ptr<float>() : nullptr,
-                     T, NB, H, W, C, Ho, Wo, F, (int)pad, (int)kchunk);
+#define LAUNCH_WGRAD(DET_)                                                     \
+  hipLaunchKernelGGL((tconv_wgrad_kernel<DET_>), grid, dim3(256), 0,           \
+                     stream.stream(),                                          \
+                     reinterpret_cast<const bf16*>(dyc.data_ptr()),            \
+                     reinterpret_cast<const bf16*>(xc.data_ptr()),             \
+                     acc.data_ptr<float>(),                                    \
+                     with_bias ? dbacc.data_ptr<float>() : nullptr,            \
+                     T, NB, H, W, C, Ho, Wo, F, (int)pad, (int)kchunk)
+  if (det) LAUNCH_WGRAD(true); else LAUNCH_WGRAD(false);
+#undef LAUNCH_WGRAD
   auto dw = torch::empty({T, F, C, 3, 3}, x.options().dtype(torch::kFloat32));
   const long total = (long)T * F * C * 9;
   hipLaunchKernelGGL(wgrad_finalize_kernel, dim3(ew_grid2(total, 256)),
                      dim3(256), 0, stream.stream(), acc.data_ptr<float>(),
-                     dw.data_ptr<float>(), T, F, C);
+                     dw.data_ptr<float>(), T, F, C, nslices);
+  auto db = nslices == 1 ? dbacc.select(1, 0).contiguous() : dbacc.sum(1);
   return {dw, db};
 }
 
