@@ -1,8 +1,8 @@
 // Fused softmax + cross-entropy for few-shot logits [T, M, ways]
 // (reference: F.cross_entropy at few_shot_learning_system.py:284).
-// Forward: one wave per row -> per-task mean loss [T] in one pass
-// (atomicAdd of row loss / M into loss[t]).  Saves softmax probs for the
-// first-order backward kernel; the create_graph path recomputes in torch.
+// Forward: one wave per row -> per-row losses, then an ordered per-task
+// mean (deterministic; no fp32 atomics).  Saves softmax probs for the
+// backward kernel; double-backward is the analytic softmax Jacobian.
 
 #include "common.h"
 #include <torch/extension.h>
@@ -13,7 +13,7 @@ using namespace maml355;
 __global__ void ce_fwd_kernel(const float* __restrict__ logits,
                               const long* __restrict__ labels,
                               float* __restrict__ probs,
-                              float* __restrict__ loss,  // [T], pre-zeroed
+                              float* __restrict__ row_loss,  // [T*M]
                               int T, int M, int ways) {
   const long rows = (long)T * M;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -29,12 +29,22 @@ __global__ void ce_fwd_kernel(const float* __restrict__ logits,
     const float seb = __shfl(se, 0, WAVE);
     if (lane < ways) probs[row * ways + lane] = e / seb;
     if (lane == 0) {
-      const long t = row / M;
       const long y = labels[row];
-      const float row_loss = logf(seb) + mxb - lrow[y];
-      atomicAdd(&loss[t], row_loss / (float)M);
+      row_loss[row] = logf(seb) + mxb - lrow[y];
     }
   }
+}
+
+// ordered per-task mean of row values [T, M] -> out [T] (deterministic —
+// the fp32-atomic version made the task loss order-dependent)
+__global__ void ce_rowmean_kernel(const float* __restrict__ rows,
+                                  float* __restrict__ out, int T, int M) {
+  const int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= T) return;
+  float s = 0.f;
+  const float* r = rows + (long)t * M;
+  for (int m = 0; m < M; ++m) s += r[m];
+  out[t] = s / (float)M;
 }
 
 __global__ void ce_bwd_kernel(const float* __restrict__ probs,
@@ -62,7 +72,7 @@ __global__ void ce_dbwd_kernel(const float* __restrict__ probs,
                                const float* __restrict__ gdl,
                                const float* __restrict__ gtask,
                                float* __restrict__ d_logits,
-                               float* __restrict__ d_gtask,  // pre-zeroed [T]
+                               float* __restrict__ d_gtask_rows,  // [T*M]
                                int T, int M, int ways) {
   const long rows = (long)T * M;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -80,7 +90,7 @@ __global__ void ce_dbwd_kernel(const float* __restrict__ probs,
     }
     if (lane == 0) {
       const float gl = gdl[row * ways + labels[row]];
-      atomicAdd(&d_gtask[t], (dotb - gl) / (float)M);
+      d_gtask_rows[row] = dotb - gl;
     }
   }
 }
@@ -98,13 +108,17 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels) {
   const int T = (int)logits.size(0), M = (int)logits.size(1),
             ways = (int)logits.size(2);
   auto probs = torch::empty_like(lf);
-  auto loss = torch::zeros({T}, lf.options());
+  auto loss = torch::empty({T}, lf.options());
+  auto rowloss = torch::empty({T, M}, lf.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   const long rows = (long)T * M;
   hipLaunchKernelGGL(ce_fwd_kernel, dim3(grid_for(rows * WAVE, 256)), dim3(256),
                      0, stream.stream(), lf.data_ptr<float>(),
                      lab.data_ptr<long>(), probs.data_ptr<float>(),
-                     loss.data_ptr<float>(), T, M, ways);
+                     rowloss.data_ptr<float>(), T, M, ways);
+  hipLaunchKernelGGL(ce_rowmean_kernel, dim3((T + 255) / 256), dim3(256), 0,
+                     stream.stream(), rowloss.data_ptr<float>(),
+                     loss.data_ptr<float>(), T, M);
   return {loss, probs};
 }
 
@@ -113,7 +127,8 @@ std::vector<torch::Tensor> ce_dbwd(torch::Tensor probs, torch::Tensor labels,
   const int T = (int)probs.size(0), M = (int)probs.size(1),
             ways = (int)probs.size(2);
   auto d_logits = torch::empty_like(probs);
-  auto d_gtask = torch::zeros({T}, probs.options());
+  auto d_gtask = torch::empty({T}, probs.options());
+  auto d_gtask_rows = torch::empty({T, M}, probs.options());
   auto lab = labels.contiguous().to(torch::kLong);
   auto gdlc = gdl.contiguous().to(torch::kFloat32);
   auto gtc = gtask.contiguous().to(torch::kFloat32);
@@ -123,7 +138,10 @@ std::vector<torch::Tensor> ce_dbwd(torch::Tensor probs, torch::Tensor labels,
                      dim3(256), 0, stream.stream(), probs.data_ptr<float>(),
                      lab.data_ptr<long>(), gdlc.data_ptr<float>(),
                      gtc.data_ptr<float>(), d_logits.data_ptr<float>(),
-                     d_gtask.data_ptr<float>(), T, M, ways);
+                     d_gtask_rows.data_ptr<float>(), T, M, ways);
+  hipLaunchKernelGGL(ce_rowmean_kernel, dim3((T + 255) / 256), dim3(256), 0,
+                     stream.stream(), d_gtask_rows.data_ptr<float>(),
+                     d_gtask.data_ptr<float>(), T, M);
   return {d_logits, d_gtask};
 }
 

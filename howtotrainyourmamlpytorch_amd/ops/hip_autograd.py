@@ -372,7 +372,8 @@ def _warn_fallback(reason: str) -> None:
 
 
 def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
-    if stride != 1 or x.dtype != torch.bfloat16 or w.shape[0] > 64 or w.shape[1] > 64:
+    # w is task-batched [T, F, C, 3, 3]: dim 1 = out-channels, dim 2 = in
+    if stride != 1 or x.dtype != torch.bfloat16 or w.shape[1] > 64 or w.shape[2] > 64:
         # stride-2 (max_pooling=False configs), fp32 compute and >64-channel
         # shapes use the grouped-ATen composition; loud, not silent
         if stride != 1:
@@ -380,7 +381,7 @@ def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
         elif x.dtype != torch.bfloat16:
             _warn_fallback(f"dtype {x.dtype}")
         else:
-            _warn_fallback(f"channels > 64 (F={w.shape[0]}, C={w.shape[1]})")
+            _warn_fallback(f"channels > 64 (F={w.shape[1]}, C={w.shape[2]})")
         wc = w.to(x.dtype)
         bc = b.to(x.dtype) if b is not None else None
         y = ref.task_conv3x3(x, wc, bc, stride, padding)

@@ -35,12 +35,12 @@ def bench_conv(T, NB, H, W, C, F, tag):
     ext = hip_ext()
     wp = ext.tconv_repack(w, False)
     wpd = ext.tconv_repack(w, True)
-    y = ext.tconv_mm(x, wp, b, 1, H, W)
+    y = ext.tconv_mm(x, wp, b, 1, H, W, False)[0]
     dy = torch.randn_like(y)
 
     flops = 2.0 * T * NB * H * W * F * 9 * C
-    t_f = timeit(lambda: ext.tconv_mm(x, wp, b, 1, H, W))
-    t_d = timeit(lambda: ext.tconv_mm(dy, wpd, None, 1, H, W))
+    t_f = timeit(lambda: ext.tconv_mm(x, wp, b, 1, H, W, False))
+    t_d = timeit(lambda: ext.tconv_mm(dy, wpd, None, 1, H, W, False))
     t_w = timeit(lambda: ext.tconv_wgrad(dy, x, 1, True))
     print(f"[conv {tag}] T{T} NB{NB} {H}x{W} C{C}->F{F}  "
           f"fwd {t_f*1e6:7.1f}us {flops/t_f/1e12:6.1f}TF | "
