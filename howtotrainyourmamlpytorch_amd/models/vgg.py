@@ -35,6 +35,26 @@ def _xavier_uniform(shape, gen: Optional[torch.Generator] = None) -> torch.Tenso
     return t
 
 
+class _SlotGather(torch.autograd.Function):
+    """Per-element LR vector from the per-slot LSLR table.
+
+    Forward is a plain gather; backward is an ORDERED per-slot segment sum
+    over the contiguous arena ranges (torch's index_select backward is an
+    atomic index_add on CUDA — run-to-run nondeterministic, measured
+    7e-7 drift on the LSLR grads).  The backward uses differentiable torch
+    ops, so create_graph (second-order MAML) works through it."""
+
+    @staticmethod
+    def forward(ctx, lrs_step, slot_index, bounds):
+        ctx.bounds = bounds
+        return lrs_step.index_select(0, slot_index)
+
+    @staticmethod
+    def backward(ctx, gout):
+        dl = torch.stack([gout[off:off + n].sum() for off, n in ctx.bounds])
+        return dl, None, None
+
+
 class TaskBatchedVGG(nn.Module):
     """4-stage conv backbone + linear head, fully functional over an arena.
 
@@ -150,8 +170,12 @@ class TaskBatchedVGG(nn.Module):
 
     def lr_vector(self, lrs: torch.Tensor, num_step: int) -> torch.Tensor:
         """Per-element learning-rate vector [P] from the LSLR table
-        [num_slots, num_steps+1] — differentiable gather."""
-        return lrs[:, num_step].index_select(0, self.slot_index)
+        [num_slots, num_steps+1] — differentiable gather with a
+        deterministic segment-sum backward."""
+        if not hasattr(self, "_slot_bounds"):
+            self._slot_bounds = [(s.offset, s.numel) for s in self.arena.specs]
+        return _SlotGather.apply(lrs[:, num_step], self.slot_index,
+                                 self._slot_bounds)
 
     # ------------------------------------------------------------------
     def forward(self, x: torch.Tensor, num_step: int, arena: torch.Tensor,

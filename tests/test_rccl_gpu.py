@@ -1,9 +1,9 @@
-"""RCCL (torch.distributed backend "nccl" on ROCm) smoke on a single
-MI355X: two ranks co-resident on one GPU (RCCL permits this) exercise the
-real collective code paths — flat-bucket all-reduce, the (count, sum,
-sumsq) statistics reduction, and the overlapped per-chunk reduction —
-so the RCCL path is hardware-tested even without an 8-GPU node
-(VERDICT round 1, item 3b)."""
+"""RCCL (torch.distributed backend "nccl" on ROCm) smoke: two ranks
+exercise the real collective code paths — flat-bucket all-reduce, the
+(count, sum, sumsq) statistics reduction, and the overlapped per-chunk
+reduction.  Measured on MI355X/RCCL 2.26: co-resident ranks on ONE device
+are REFUSED ("Duplicate GPU detected"), so this needs >= 2 GPUs and runs
+when the driver has a multi-GPU node; single-GPU boxes skip."""
 
 import os
 
@@ -18,12 +18,11 @@ def _nccl_worker(rank, world, rdv_file, out_dir):
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     import torch.distributed as dist
-    # both ranks share cuda:0 — RCCL supports co-resident ranks
-    torch.cuda.set_device(0)
+    torch.cuda.set_device(rank)
     dist.init_process_group("nccl", init_method=f"file://{rdv_file}",
                             rank=rank, world_size=world)
     from howtotrainyourmamlpytorch_amd.parallel.dist import DistContext
-    ctx = DistContext(rank, world, 0, "nccl")
+    ctx = DistContext(rank, world, rank, "nccl")
 
     # flat-bucket all-reduce of parameter gradients
     torch.manual_seed(10 + rank)
@@ -59,7 +58,10 @@ def _nccl_worker(rank, world, rdv_file, out_dir):
     dist.destroy_process_group()
 
 
-def test_rccl_two_ranks_one_gpu(tmp_path):
+def test_rccl_two_ranks(tmp_path):
+    if torch.cuda.device_count() < 2:
+        pytest.skip("RCCL refuses co-resident ranks on one device "
+                    "(verified on MI355X, RCCL 2.26); needs >= 2 GPUs")
     rdv = str(tmp_path / "rdv_nccl")
     ctx = mp.get_context("spawn")
     procs = []
