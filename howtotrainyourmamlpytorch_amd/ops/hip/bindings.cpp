@@ -43,6 +43,11 @@ std::vector<torch::Tensor> tconv_mm(torch::Tensor x, torch::Tensor wp,
                                     long Ho, long Wo, bool with_stats);
 std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
                                        long pad, bool with_bias);
+torch::Tensor tconv_repack_v2(torch::Tensor w, bool dgrad);
+std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
+                                       c10::optional<torch::Tensor> bias,
+                                       long pad, long Ho, long Wo, long Co,
+                                       bool with_stats);
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B);
 // adam.hip
 void adam_step(std::vector<torch::Tensor> params,
@@ -68,6 +73,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tconv_repack", &tconv_repack, "repack conv weights for MFMA (fwd/dgrad)");
   m.def("tconv_mm", &tconv_mm, "task-batched MFMA 3x3 conv fwd/dgrad");
   m.def("tconv_wgrad", &tconv_wgrad, "task-batched MFMA 3x3 conv wgrad");
+  m.def("tconv_repack_v2", &tconv_repack_v2,
+        "repack conv weights into the v2 swizzled LDS image");
+  m.def("tconv_mm_v2", &tconv_mm_v2,
+        "async-pipelined MFMA 3x3 conv fwd/dgrad (v2)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam + grad clamp");
 }

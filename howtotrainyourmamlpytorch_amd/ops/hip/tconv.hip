@@ -301,6 +301,248 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 }
 
 // ---------------------------------------------------------------------------
+// v2 fwd/dgrad: async-staged 2-phase double-buffered schedule.
+//
+// The v1 kernel stages tiles synchronously through registers between two
+// barriers — measured VALU/barrier-bound at 4-8% of bf16 peak (round-1
+// profiles).  v2 restructures per the platform guide's pipeline catalog
+// (§5.5 T3 minimum-2-phase): while the MFMAs of K-step k run, the loads
+// of K-step k+1 are already in flight via `global_load_lds` (async
+// global->LDS DMA, 16B/lane, no register round-trip, no VALU staging):
+//
+//   prologue: STAGE(buf0, ks=0); barrier
+//   loop ks:  STAGE(buf^1, ks+1)   // issue async loads, no wait
+//             ds_read + MFMA from buf
+//             barrier (drains vmcnt -> next tile ready); buf ^= 1
+//
+// Enablers (all pre-computed on the host side):
+//   * input is PADDED ([T,NB,H+2p,W+2p,Ci]) so every im2col address is
+//     valid -> no boundary predicates, per-lane source addresses only
+//   * weights are repacked into the exact swizzled LDS image
+//     ([T,ksteps,64,BK] with the XOR swizzle pre-applied and zeros in
+//     the K/Co tails) -> B staging is a linear byte copy
+//   * A staging pre-swizzles the SOURCE column slot (j ^= m&7, guide
+//     ERRATA 21: global_load_lds writes linearly, so the swizzle must
+//     live in the source address; the ds_read side applies the same
+//     involution)
+//   * K-tail / M-tail lanes read a 16B zero page instead of predicating
+// Requires Ci % 8 == 0 (the first conv layer's C in {1,3} keeps v1).
+// LDS: 2x(256x64) A + 2x(64x64) B bf16 = 80 KB dynamic -> 2 blocks/CU.
+// ---------------------------------------------------------------------------
+#define V2_LDS_BYTES (2 * BM * BK * 2 + 2 * 64 * BK * 2)
+
+__global__ __launch_bounds__(512, 2)
+void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wimg,
+                        const float* __restrict__ bias, const bf16* __restrict__ zpage,
+                        bf16* __restrict__ Y, float* __restrict__ sums_out,
+                        int T, int NB, int Hp, int Wp, int Ci,
+                        int Ho, int Wo, int Co) {
+  const int t = blockIdx.y;
+  const long Mtot = (long)NB * Ho * Wo;
+  const long m0 = (long)blockIdx.x * BM;
+  const int K9 = 9 * Ci;
+  const int ksteps = (K9 + BK - 1) / BK;
+  const int ntiles = (Co + 15) / 16;
+
+  extern __shared__ short smem[];
+  // layout: A buffers at [0, BM*BK), [BM*BK, 2*BM*BK); B buffers after
+#define LDS_A(buf_) (smem + (buf_) * (BM * BK))
+#define LDS_B(buf_) (smem + 2 * (BM * BK) + (buf_) * (64 * BK))
+
+  const bf16* Xt = Xp + (long)t * NB * Hp * Wp * Ci;
+  const short* Wt = (const short*)Wimg + (long)t * ksteps * 64 * BK;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  // loop-invariant per-thread A-staging descriptors: 4 slots of 16B; slot
+  // s -> (row m = s>>3, dest 8-col group j = s&7); source col group is
+  // j ^ (m&7) (inverse swizzle on the source side)
+  int a_rb[4];   // element offset of (n, ho, wo) in the padded input, or -1
+  int a_k8[4];   // source k base = (j ^ (m&7)) * 8 within the K-step
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const int s = q * 512 + threadIdx.x;
+    const int m = s >> 3;
+    a_k8[q] = ((s & 7) ^ (m & 7)) * 8;
+    const long mg = m0 + m;
+    if (mg < Mtot) {
+      const int wo = (int)(mg % Wo);
+      const int ho = (int)((mg / Wo) % Ho);
+      const int n = (int)(mg / ((long)Wo * Ho));
+      a_rb[q] = ((n * Hp + ho) * Wp + wo) * Ci;
+    } else {
+      a_rb[q] = -1;
+    }
+  }
+
+#define V2_STAGE(ks_, buf_)                                                    \
+  do {                                                                         \
+    const int k0_ = (ks_)*BK;                                                  \
+    _Pragma("unroll")                                                          \
+    for (int q = 0; q < 4; ++q) {                                              \
+      const int k = k0_ + a_k8[q];                                             \
+      const bf16* src;                                                         \
+      if (a_rb[q] < 0 || k >= K9) {                                            \
+        src = zpage;                                                           \
+      } else {                                                                 \
+        const int kyx = k / Ci;                                                \
+        const int c = k - kyx * Ci;                                            \
+        src = Xt + a_rb[q] + ((kyx / 3) * Wp + (kyx % 3)) * Ci + c;            \
+      }                                                                        \
+      auto ldst = (__attribute__((address_space(3))) void*)(                   \
+          &LDS_A(buf_)[((long)q * 512 + wave * 64) * 8]);                      \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) void*)src, ldst, 16, 0, 0); \
+    }                                                                          \
+    {                                                                          \
+      const short* bsrc = Wt + (long)(ks_)*64 * BK + (long)threadIdx.x * 8;    \
+      auto ldst = (__attribute__((address_space(3))) void*)(                   \
+          &LDS_B(buf_)[(long)wave * 64 * 8]);                                  \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) void*)bsrc, ldst, 16, 0, 0);\
+    }                                                                          \
+  } while (0)
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[hh][i] = {0.f, 0.f, 0.f, 0.f};
+
+  V2_STAGE(0, 0);
+  __syncthreads();
+
+  int buf = 0;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    if (ks + 1 < ksteps) V2_STAGE(ks + 1, buf ^ 1);
+    const short* la = LDS_A(buf);
+    const short* lb = LDS_B(buf);
+#pragma unroll
+    for (int ksl = 0; ksl < BK / 32; ++ksl) {
+      bf16x8 a0 = *(const bf16x8*)&la[swz64(wave * 32 + fr, ksl * 32 + fk * 8)];
+      bf16x8 a1 = *(const bf16x8*)&la[swz64(wave * 32 + 16 + fr, ksl * 32 + fk * 8)];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        if (nt < ntiles) {
+          bf16x8 b = *(const bf16x8*)&lb[swz64(nt * 16 + fr, ksl * 32 + fk * 8)];
+          acc[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc[0][nt], 0, 0, 0);
+          acc[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, acc[1][nt], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();  // drains vmcnt -> buf^1 staged; all reads of buf done
+    buf ^= 1;
+  }
+#undef V2_STAGE
+
+  // epilogue — identical to v1
+  __shared__ float sums_lds[2][64];
+  if (sums_out) {
+    for (int i = threadIdx.x; i < 128; i += blockDim.x) {
+      sums_lds[i >> 6][i & 63] = 0.f;
+    }
+    __syncthreads();
+  }
+  bf16* Yt = Y + (long)t * Mtot * Co;
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+  for (int nt = 0; nt < 4; ++nt) {
+    if (nt >= ntiles) break;
+    const int col = nt * 16 + fr;
+    if (col >= Co) continue;
+    const float bv = bias ? bias[(long)t * Co + col] : 0.f;
+    float ls = 0.f, lq = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int row = wave * 32 + hh * 16 + fk * 4 + j;
+      const long mg = m0 + row;
+      if (mg < Mtot) {
+        const float v = acc[hh][nt][j] + bv;
+        ((short*)Yt)[mg * Co + col] =
+            (short)__bfloat16_as_short(__float2bfloat16(v));
+        ls += v;
+        lq += v * v;
+      }
+    }
+    if (sums_out) {
+      atomicAdd(&sums_lds[0][col], ls);
+      atomicAdd(&sums_lds[1][col], lq);
+    }
+  }
+  if (sums_out) {
+    __syncthreads();
+    for (int c = threadIdx.x; c < Co; c += blockDim.x) {
+      atomicAdd(&sums_out[((long)t * 2 + 0) * Co + c], sums_lds[0][c]);
+      atomicAdd(&sums_out[((long)t * 2 + 1) * Co + c], sums_lds[1][c]);
+    }
+  }
+}
+
+#undef LDS_A
+#undef LDS_B
+
+// NHWC zero-pad: [T, NB, H, W, C] -> [T, NB, H+2p, W+2p, C] (C % 8 == 0)
+__global__ void pad_nhwc_kernel(const bf16* __restrict__ x, bf16* __restrict__ xp,
+                                int T, int NB, int H, int W, int C, int pad) {
+  const int Hp = H + 2 * pad, Wp = W + 2 * pad;
+  const int c8n = C / 8;
+  const long total = (long)T * NB * Hp * Wp * c8n;
+  typedef bf16x8 v8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    const int c0 = (int)(i % c8n) * 8;
+    long r = i / c8n;
+    const int wp = (int)(r % Wp); r /= Wp;
+    const int hp = (int)(r % Hp); r /= Hp;
+    const long nb = r;  // t*NB + nb combined
+    const int h = hp - pad, w = wp - pad;
+    v8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (h >= 0 && h < H && w >= 0 && w < W) {
+      v = *(const v8*)&((const short*)x)[((nb * H + h) * W + w) * C + c0];
+    }
+    *(v8*)&((short*)xp)[((nb * Hp + hp) * Wp + wp) * C + c0] = v;
+  }
+}
+
+// Weight repack for v2: W [T, F, C, 3, 3] fp32 -> the exact swizzled LDS
+// B-image [T, ksteps, 64, BK] bf16 consumed by tconv_mm_v2_kernel
+// (zeros in the K9/Co tails; XOR swizzle pre-applied per ERRATA 21).
+__global__ void repack_v2_kernel(const float* __restrict__ w,
+                                 bf16* __restrict__ out,
+                                 int T, int F, int C, int ksteps, bool dgrad) {
+  const int cin = dgrad ? F : C;   // GEMM inner-channel count
+  const int cout = dgrad ? C : F;  // GEMM output columns
+  const int K9 = 9 * cin;
+  const long total = (long)T * ksteps * 64 * BK;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    const int jj = (int)(i & 7);
+    const int j = (int)((i >> 3) & 7);
+    const int f = (int)((i >> 6) & 63);
+    long r = i >> 12;
+    const int ks = (int)(r % ksteps);
+    const long t = r / ksteps;
+    const int k = ks * BK + ((j ^ (f & 7)) * 8) + jj;
+    float v = 0.f;
+    if (k < K9 && f < cout) {
+      const int kyx = k / cin;
+      const int ci = k - kyx * cin;
+      const int ky = kyx / 3, kx = kyx % 3;
+      if (!dgrad) {
+        v = w[((((long)t * F + f) * C + ci) * 3 + ky) * 3 + kx];
+      } else {
+        v = w[((((long)t * F + ci) * C + f) * 3 + (2 - ky)) * 3 + (2 - kx)];
+      }
+    }
+    out[i] = __float2bfloat16(v);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // wgrad kernel: dWacc[t][n=9C][f] += sum_k dY[t,k,f] * im2col(X)[t,k,n]
 // Block: 256 thr = 4 waves; wave w owns n-subtile w (16 cols), iterates
 // m-tiles over F.  K-chunked grid with fp32 atomicAdd.
@@ -575,6 +817,81 @@ std::vector<torch::Tensor> tconv_mm(torch::Tensor x, torch::Tensor wp,
                      reinterpret_cast<bf16*>(y.data_ptr()),
                      with_stats ? sums.data_ptr<float>() : nullptr,
                      T, NB, H, W, Ci, (int)Ho, (int)Wo, Co, (int)pad);
+  return {y, sums};
+}
+
+torch::Tensor tconv_repack_v2(torch::Tensor w, bool dgrad) {
+  TORCH_CHECK(w.is_cuda() && w.dim() == 5 && w.size(3) == 3 && w.size(4) == 3);
+  auto wc = w.contiguous().to(torch::kFloat32);
+  const int T = (int)w.size(0), F = (int)w.size(1), C = (int)w.size(2);
+  const int cin = dgrad ? F : C;
+  const int ksteps = (9 * cin + BK - 1) / BK;
+  auto out = torch::empty({T, ksteps, 64, BK},
+                          w.options().dtype(torch::kBFloat16));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const long total = (long)T * ksteps * 64 * BK;
+  hipLaunchKernelGGL(repack_v2_kernel, dim3(ew_grid2(total, 256)), dim3(256), 0,
+                     stream.stream(), wc.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(out.data_ptr()), T, F, C, ksteps,
+                     dgrad);
+  return out;
+}
+
+// v2 conv: x UNPADDED [T, NB, H, W, Ci] bf16 (Ci % 8 == 0), wimg from
+// tconv_repack_v2 (Co zero-padded to 64 rows, so the caller passes the
+// true Co); pads internally; returns {y, sums}
+std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
+                                       c10::optional<torch::Tensor> bias,
+                                       long pad, long Ho, long Wo, long Co,
+                                       bool with_stats) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 5 && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "tconv_mm_v2 needs bf16");
+  const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3), Ci = (int)x.size(4);
+  TORCH_CHECK(Ci % 8 == 0 && Co <= 64);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  // pad (p == 0 -> use x directly: every im2col address is already valid)
+  torch::Tensor xp = x;
+  int Hp = H, Wp = W;
+  if (pad > 0) {
+    Hp = H + 2 * (int)pad;
+    Wp = W + 2 * (int)pad;
+    xp = torch::empty({T, NB, Hp, Wp, Ci}, x.options());
+    const long ptotal = (long)T * NB * Hp * Wp * (Ci / 8);
+    hipLaunchKernelGGL(pad_nhwc_kernel, dim3(ew_grid2(ptotal, 256)), dim3(256),
+                       0, stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<bf16*>(xp.data_ptr()),
+                       T, NB, H, W, Ci, (int)pad);
+  }
+  auto y = torch::empty({T, NB, Ho, Wo, Co}, x.options());
+  auto sums = with_stats
+                  ? torch::zeros({T, 2, Co}, x.options().dtype(torch::kFloat32))
+                  : torch::empty({0}, x.options().dtype(torch::kFloat32));
+  auto zpage_cache = torch::zeros({16}, x.options());
+  const float* bptr = nullptr;
+  torch::Tensor bc;
+  if (bias.has_value()) {
+    bc = bias->contiguous().to(torch::kFloat32);
+    bptr = bc.data_ptr<float>();
+  }
+  const long Mtot = (long)NB * Ho * Wo;
+  dim3 grid((unsigned)((Mtot + BM - 1) / BM), T);
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)tconv_mm_v2_kernel,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        V2_LDS_BYTES);
+    attr_set = true;
+  }
+  hipLaunchKernelGGL(tconv_mm_v2_kernel, grid, dim3(512), V2_LDS_BYTES,
+                     stream.stream(),
+                     reinterpret_cast<const bf16*>(xp.data_ptr()),
+                     reinterpret_cast<const bf16*>(wimg.data_ptr()), bptr,
+                     reinterpret_cast<const bf16*>(zpage_cache.data_ptr()),
+                     reinterpret_cast<bf16*>(y.data_ptr()),
+                     with_stats ? sums.data_ptr<float>() : nullptr,
+                     T, NB, Hp, Wp, Ci, (int)Ho, (int)Wo, (int)Co);
   return {y, sums};
 }
 

@@ -268,6 +268,14 @@ def lslr_update(arena, grad, lr_vec):
 # custom kernels at EVERY derivative order (second-order MAML's
 # create_graph included), no torch fallback on the hot path.
 # ---------------------------------------------------------------------------
+def _conv_v2_ok(ci: int) -> bool:
+    """v2 (async global_load_lds pipeline) needs 8-aligned input channels.
+    Measured SLOWER than v1 at the flagship shapes (156 vs 193 TF conv1:
+    80 KB LDS drops occupancy 3->2 blocks/CU and the in-loop k-offset
+    divisions add VALU) — opt-in via MAML355_CONV_V2=1 until it wins."""
+    return ci % 8 == 0 and os.environ.get("MAML355_CONV_V2", "0") == "1"
+
+
 class _ConvFwdFn(torch.autograd.Function):
     """Returns (y, bn_sums): with want_stats the epilogue accumulates the
     following BN's per-channel sum/sum-of-squares for free."""
@@ -277,10 +285,15 @@ class _ConvFwdFn(torch.autograd.Function):
         ctx.save_for_backward(x, w)
         ctx.pad = pad
         ctx.has_bias = b is not None
-        wp = _ext().tconv_repack(w, False)
         H, W = x.shape[2], x.shape[3]
         Ho, Wo = H + 2 * pad - 2, W + 2 * pad - 2
-        y, sums = _ext().tconv_mm(x, wp, b, pad, Ho, Wo, want_stats)
+        if _conv_v2_ok(x.shape[4]):
+            wp = _ext().tconv_repack_v2(w, False)
+            y, sums = _ext().tconv_mm_v2(x, wp, b, pad, Ho, Wo, w.shape[1],
+                                         want_stats)
+        else:
+            wp = _ext().tconv_repack(w, False)
+            y, sums = _ext().tconv_mm(x, wp, b, pad, Ho, Wo, want_stats)
         ctx.mark_non_differentiable(sums)
         return y, sums
 
@@ -313,9 +326,13 @@ class _ConvDgradFn(torch.autograd.Function):
     def forward(ctx, dy, w, pad):
         ctx.save_for_backward(dy, w)
         ctx.pad = pad
-        wp = _ext().tconv_repack(w, True)
         Ho, Wo = dy.shape[2], dy.shape[3]
         H, W = Ho - 2 * pad + 2, Wo - 2 * pad + 2
+        if _conv_v2_ok(dy.shape[4]):
+            wp = _ext().tconv_repack_v2(w, True)
+            return _ext().tconv_mm_v2(dy, wp, None, 2 - pad, H, W,
+                                      w.shape[2], False)[0]
+        wp = _ext().tconv_repack(w, True)
         return _ext().tconv_mm(dy, wp, None, 2 - pad, H, W, False)[0]
 
     @staticmethod

@@ -105,3 +105,34 @@ def test_tconv_wgrad_large_k_split():
     (dwr,) = torch.autograd.grad(yr, (wr,), gout.float().cpu())
     scale = dwr.abs().max().item()
     torch.testing.assert_close(dw.cpu(), dwr, rtol=5e-2, atol=3e-2 * max(scale, 1.0))
+
+
+def test_conv_v2_bitwise_matches_v1():
+    """The v2 async-pipelined kernel performs the identical MFMA sequence
+    (same tiles, zero tails), so its output is bitwise equal to v1."""
+    from howtotrainyourmamlpytorch_amd.ops import hip_ext
+    ext = hip_ext()
+    torch.manual_seed(11)
+    for (T, NB, H, W, C, F, pad) in [(3, 7, 14, 14, 48, 48, 1),
+                                     (2, 9, 28, 28, 64, 64, 1),
+                                     (2, 5, 12, 12, 16, 48, 0),
+                                     (1, 3, 10, 10, 8, 24, 1)]:
+        x = torch.randn(T, NB, H, W, C, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(T, F, C, 3, 3, device="cuda")
+        b = torch.randn(T, F, device="cuda")
+        Ho, Wo = H + 2 * pad - 2, W + 2 * pad - 2
+        wp1 = ext.tconv_repack(w, False)
+        y1 = ext.tconv_mm(x, wp1, b, pad, Ho, Wo, False)[0]
+        wp2 = ext.tconv_repack_v2(w, False)
+        y2 = ext.tconv_mm_v2(x, wp2, b, pad, Ho, Wo, F, False)[0]
+        assert (y1 == y2).all().item(), \
+            f"v2 != v1 at {(T, NB, H, W, C, F, pad)}: " \
+            f"max {(y1.float() - y2.float()).abs().max().item():e}"
+        # dgrad orientation (flipped repack, pad' = 2 - pad)
+        dy = torch.randn(T, NB, Ho, Wo, F, device="cuda", dtype=torch.bfloat16)
+        wpd1 = ext.tconv_repack(w, True)
+        dx1 = ext.tconv_mm(dy, wpd1, None, 2 - pad, H, W, False)[0]
+        wpd2 = ext.tconv_repack_v2(w, True)
+        dx2 = ext.tconv_mm_v2(dy, wpd2, None, 2 - pad, H, W, C, False)[0]
+        assert (dx1 == dx2).all().item(), \
+            f"v2 dgrad != v1 at {(T, NB, H, W, C, F, pad)}"

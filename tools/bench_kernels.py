@@ -46,6 +46,14 @@ def bench_conv(T, NB, H, W, C, F, tag):
           f"fwd {t_f*1e6:7.1f}us {flops/t_f/1e12:6.1f}TF | "
           f"dgrad {t_d*1e6:7.1f}us {flops/t_d/1e12:6.1f}TF | "
           f"wgrad {t_w*1e6:7.1f}us {flops/t_w/1e12:6.1f}TF")
+    if C % 8 == 0:
+        wp2 = ext.tconv_repack_v2(w, False)
+        wpd2 = ext.tconv_repack_v2(w, True)
+        t_f2 = timeit(lambda: ext.tconv_mm_v2(x, wp2, b, 1, H, W, F, False))
+        t_d2 = timeit(lambda: ext.tconv_mm_v2(dy, wpd2, None, 1, H, W, C, False))
+        print(f"[conv {tag}]   v2:                 "
+              f"fwd {t_f2*1e6:7.1f}us {flops/t_f2/1e12:6.1f}TF | "
+              f"dgrad {t_d2*1e6:7.1f}us {flops/t_d2/1e12:6.1f}TF")
 
 
 def bench_bn(T, M, C, tag):
