@@ -44,6 +44,8 @@ std::vector<torch::Tensor> tconv_mm(torch::Tensor x, torch::Tensor wp,
 std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
                                        long pad, bool with_bias);
 torch::Tensor tconv_repack_v2(torch::Tensor w, bool dgrad);
+std::vector<torch::Tensor> tconv_wgrad_v2(torch::Tensor dy, torch::Tensor x,
+                                          long pad, bool with_bias);
 std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
                                        c10::optional<torch::Tensor> bias,
                                        long pad, long Ho, long Wo, long Co,
@@ -77,6 +79,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "repack conv weights into the v2 swizzled LDS image");
   m.def("tconv_mm_v2", &tconv_mm_v2,
         "async-pipelined MFMA 3x3 conv fwd/dgrad (v2)");
+  m.def("tconv_wgrad_v2", &tconv_wgrad_v2,
+        "async-pipelined wgrad via k-contiguous operand transposes (v2)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam + grad clamp");
 }

@@ -112,6 +112,10 @@ __global__ void wgrad_finalize_kernel(const float* __restrict__ acc,
 // ---------------------------------------------------------------------------
 #define BM 256
 #define BK 64    // K-step: 2 MFMA K-slices per barrier
+#define WG_KCHUNK 4096
+#define WBK 64   // wgrad K-step (2 MFMA K-slices per barrier)
+
+#define WGN 64   // wgrad output columns per block (4 waves x 16)
 
 __global__ __launch_bounds__(512, 2)
 void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
@@ -543,15 +547,321 @@ __global__ void repack_v2_kernel(const float* __restrict__ w,
 }
 
 // ---------------------------------------------------------------------------
+// wgrad v2 support: global transposes that make BOTH wgrad operands
+// k-contiguous, so the main kernel stages tiles with linear async
+// `global_load_lds` instead of v1's 8-scalar-LDS-write transposes.
+//
+//   dYT [T, F, Kr]: dY transposed to [f][k'] with each (n, ho) line padded
+//     to Wo8 = roundup(Wo, 8) columns (zeros in the pad -> those k
+//     contribute nothing), k' = (n*Ho + ho)*Wo8 + wo, Kr = roundup(.., 64).
+//   XT  [T, C, NB, Hxp, Wxp]: X transposed channel-major and zero-padded
+//     (Hxp = H + 2p, Wxp = Wo8 + 2) so the im2col B-operand read for
+//     column n=(c,dy,dx) at position k=(nimg,ho,wo) is the contiguous run
+//     XT[c][nimg][ho+dy][wo+dx ..+7] — valid for EVERY C (including the
+//     first layer's C in {1,3}, which v1 staged scalar).
+// ---------------------------------------------------------------------------
+
+// dY [T,NB,Ho,Wo,F] -> dYT [T,F,Kr]; one block per (t,n,ho) line, tiled
+// in 64-wide w-chunks (any Wo).
+__global__ void dyt_pad_kernel(const bf16* __restrict__ dy,
+                               bf16* __restrict__ dyt,
+                               int T, int NB, int Ho, int Wo, int F,
+                               int Wo8, long Kr) {
+  __shared__ short lds[64 * 72];  // [f][wo] tile, stride 72 (16B-aligned)
+  long b = blockIdx.x;
+  const int ho = (int)(b % Ho); b /= Ho;
+  const int n = (int)(b % NB); b /= NB;
+  const int t = (int)b;
+  const int f8n = (F + 7) / 8;
+  const short* src = (const short*)dy + ((((long)t * NB + n) * Ho + ho) * Wo) * F;
+  const long k0 = ((long)n * Ho + ho) * Wo8;
+  short* dst = (short*)dyt + (long)t * F * Kr + k0;
+  for (int wbase = 0; wbase < Wo8; wbase += 64) {
+    const int wlen = min(64, Wo8 - wbase);
+    for (int s = threadIdx.x; s < wlen * f8n; s += blockDim.x) {
+      const int wl = s / f8n;
+      const int wo = wbase + wl;
+      const int f0 = (s % f8n) * 8;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (wo < Wo && f0 + 8 <= F) {
+        v = *(const bf16x8*)&src[(long)wo * F + f0];
+      } else if (wo < Wo) {
+        for (int j = 0; j < 8 && f0 + j < F; ++j)
+          v[j] = src[(long)wo * F + f0 + j];
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds[(f0 + j) * 72 + wl] = v[j];
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < F * (wlen / 8); s += blockDim.x) {
+      const int f = s / (wlen / 8);
+      const int w0 = (s % (wlen / 8)) * 8;
+      *(bf16x8*)&dst[(long)f * Kr + wbase + w0] =
+          *(const bf16x8*)&lds[f * 72 + w0];
+    }
+    __syncthreads();
+  }
+}
+
+// zero dYT's K' .. Kr tail (< 64 columns per row)
+__global__ void dyt_tail_kernel(bf16* __restrict__ dyt, int T, int F,
+                                long Kprime, long Kr) {
+  const long tail = Kr - Kprime;
+  const long total = (long)T * F * tail;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += grid_stride()) {
+    const long c = i % tail;
+    const long r = i / tail;
+    ((short*)dyt)[r * Kr + Kprime + c] = 0;
+  }
+}
+
+// X [T,NB,H,W,C] -> XT [T,C,NB,Hxp,Wxp]; block per (t, n, hxp, w-tile).
+__global__ void xt_pad_kernel(const bf16* __restrict__ x, bf16* __restrict__ xt,
+                              int T, int NB, int H, int W, int C,
+                              int Hxp, int Wxp, int pad) {
+  __shared__ short lds[64 * 72];  // [c][wx] tile
+  long b = blockIdx.x;
+  const int wt = (int)(b % ((Wxp + 63) / 64)); b /= (Wxp + 63) / 64;
+  const int hxp = (int)(b % Hxp); b /= Hxp;
+  const int n = (int)(b % NB); b /= NB;
+  const int t = (int)b;
+  const int h = hxp - pad;
+  const int w0 = wt * 64;
+  const int c8n = (C + 7) / 8;
+  const short* src = (const short*)x + (((long)t * NB + n) * H) * W * C;
+  for (int s = threadIdx.x; s < 64 * c8n; s += blockDim.x) {
+    const int wl = s / c8n;
+    const int c0 = (s % c8n) * 8;
+    const int w = w0 + wl - pad;
+    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (h >= 0 && h < H && w >= 0 && w < W) {
+      if (c0 + 8 <= C) {
+        v = *(const bf16x8*)&src[((long)h * W + w) * C + c0];
+      } else {
+        for (int j = 0; j < 8 && c0 + j < C; ++j)
+          v[j] = src[((long)h * W + w) * C + c0 + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds[(c0 + j) * 72 + wl] = v[j];
+  }
+  __syncthreads();
+  short* dst = (short*)xt + ((((long)t * C) * NB + n) * Hxp + hxp) * Wxp;
+  for (int s = threadIdx.x; s < C * 8; s += blockDim.x) {
+    const int c = s / 8;
+    const int wg = (s % 8) * 8;
+    const int wx = w0 + wg;
+    if (wx + 8 <= Wxp) {
+      *(bf16x8*)&dst[((long)c * NB * Hxp) * Wxp + wx] =
+          *(const bf16x8*)&lds[c * 72 + wg];
+    } else {
+      for (int j = 0; j < 8 && wx + j < Wxp; ++j)
+        dst[((long)c * NB * Hxp) * Wxp + wx + j] = lds[c * 72 + wg + j];
+    }
+  }
+}
+
+// async-staged wgrad main kernel: A = dYT rows (f), B = im2col columns
+// from XT; 4 waves, double-buffered tiles, 2-phase schedule.  NSUB = how
+// many 64-wide n-column blocks this block owns (1 or 2): NSUB=2 doubles
+// the MFMA work amortizing each barrier at 2x the B-tile LDS.
+template <int NSUB, bool DET>
+__global__ __launch_bounds__(256, 4)
+void tconv_wgrad_v2_kernel(const bf16* __restrict__ dYT,
+                           const bf16* __restrict__ XT,
+                           const bf16* __restrict__ zpage,
+                           float* __restrict__ dWacc,  // [T, nsl, 9C, F]
+                           float* __restrict__ dBacc,  // [T, nsl, F] or null
+                           int T, int NB, int Ho, int Wo8, int Hxp, int Wxp,
+                           int C, int F, long Kr, long Kprime, int kchunk) {
+  const int t = blockIdx.z;
+  const int n0 = blockIdx.x * (64 * NSUB);
+  const int N9 = 9 * C;
+  const long kc0 = (long)blockIdx.y * kchunk;
+  const long kc_end = min(kc0 + (long)kchunk, Kr);
+  const int mtiles = (F + 15) / 16;
+  const long imgS = (long)NB * Hxp * Wxp;
+
+  __shared__ short lds_at[2][64 * WBK];
+  __shared__ short lds_bt[2][64 * NSUB * WBK];
+
+  const bf16* dYTt = dYT + (long)t * F * Kr;
+  const bf16* XTt = XT + (long)t * C * imgS;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  // ---- per-thread staging descriptors (2 A slots + 2 B slots) ----
+  // A slot s: row f = s>>3, dest col group j = s&7, src group j^(f&7)
+  int a_f[2], a_k8[2];
+#pragma unroll
+  for (int q = 0; q < 2; ++q) {
+    const int s = q * 256 + threadIdx.x;
+    a_f[q] = s >> 3;
+    a_k8[q] = ((s & 7) ^ ((s >> 3) & 7)) * 8;
+  }
+  // B slot s: col n = s>>3 (global n0+n), src k group (s&7)^(n&7);
+  // incremental k -> (image row offset, wo8) state per slot so the main
+  // loop advances with adds/compares only (no division)
+  long b_imgoff[2 * NSUB];  // c*imgS + ((nimg*Hxp)+ho+dy)*Wxp + dx
+  int b_wo8[2 * NSUB], b_ho[2 * NSUB];
+  long b_k[2 * NSUB];
+  bool b_valid[2 * NSUB];
+#pragma unroll
+  for (int q = 0; q < 2 * NSUB; ++q) {
+    const int s = q * 256 + threadIdx.x;
+    const int n = s >> 3;
+    const int ng = n0 + n;
+    b_valid[q] = ng < N9;
+    const int kyx = b_valid[q] ? ng / C : 0;
+    const int c = ng - kyx * C;
+    b_k[q] = kc0 + ((s & 7) ^ (n & 7)) * 8;
+    const int wo8 = (int)(b_k[q] % Wo8);
+    const long line = b_k[q] / Wo8;
+    const int ho = (int)(line % Ho);
+    const int nimg = (int)(line / Ho);
+    b_wo8[q] = wo8;
+    b_ho[q] = ho;
+    b_imgoff[q] = ((long)c * NB + nimg) * Hxp * Wxp +
+                  (long)(ho + kyx / 3) * Wxp + kyx % 3;
+  }
+
+#define WG2_STAGE(k0_, buf_)                                                   \
+  do {                                                                         \
+    _Pragma("unroll")                                                          \
+    for (int q = 0; q < 2; ++q) {                                              \
+      const bf16* src = (a_f[q] < F)                                           \
+                            ? dYTt + (long)a_f[q] * Kr + (k0_) + a_k8[q]       \
+                            : zpage;                                           \
+      auto ldst = (__attribute__((address_space(3))) void*)(                   \
+          &lds_at[buf_][((long)q * 256 + wave * 64) * 8]);                     \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) void*)src, ldst, 16, 0, 0); \
+    }                                                                          \
+    _Pragma("unroll")                                                          \
+    for (int q = 0; q < 2 * NSUB; ++q) {                                       \
+      const bf16* src = (b_valid[q] && b_k[q] < Kprime)                        \
+                            ? XTt + b_imgoff[q] + b_wo8[q]                     \
+                            : zpage;                                           \
+      auto ldst = (__attribute__((address_space(3))) void*)(                   \
+          &lds_bt[buf_][((long)q * 256 + wave * 64) * 8]);                     \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) void*)src, ldst, 16, 0, 0); \
+    }                                                                          \
+  } while (0)
+
+  // advance the B k-state by one K-step (WBK columns)
+#define WG2_ADVANCE()                                                          \
+  do {                                                                         \
+    _Pragma("unroll")                                                          \
+    for (int q = 0; q < 2 * NSUB; ++q) {                                       \
+      b_k[q] += WBK;                                                           \
+      b_wo8[q] += WBK;                                                         \
+      while (b_wo8[q] >= Wo8) {                                                \
+        b_wo8[q] -= Wo8;                                                       \
+        ++b_ho[q];                                                             \
+        b_imgoff[q] += Wxp;                                                    \
+        if (b_ho[q] == Ho) {                                                   \
+          b_ho[q] = 0;                                                         \
+          b_imgoff[q] += (long)(Hxp - Ho) * Wxp;                               \
+        }                                                                      \
+      }                                                                        \
+    }                                                                          \
+  } while (0)
+
+  f32x4 acc[NSUB][4];
+#pragma unroll
+  for (int sb = 0; sb < NSUB; ++sb)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[sb][i] = {0.f, 0.f, 0.f, 0.f};
+  const bool do_bias = (dBacc != nullptr) && (blockIdx.x == 0);
+  float db_acc = 0.f;
+  const int db_f = threadIdx.x & 63;
+  const int db_q = threadIdx.x >> 6;
+
+  WG2_STAGE(kc0, 0);
+  __syncthreads();
+  int buf = 0;
+  for (long k0 = kc0; k0 < kc_end; k0 += WBK) {
+    WG2_ADVANCE();
+    if (k0 + WBK < kc_end) WG2_STAGE(k0 + WBK, buf ^ 1);
+    const short* la = lds_at[buf];
+    const short* lb = lds_bt[buf];
+#pragma unroll
+    for (int ks = 0; ks < WBK / 32; ++ks) {
+#pragma unroll
+      for (int sb = 0; sb < NSUB; ++sb) {
+        // wave's n-subtile within B rows: (sb*4 + wave)*16
+        bf16x8 bfrag = *(const bf16x8*)&lb[swz64((sb * 4 + wave) * 16 + fr,
+                                                 ks * 32 + fk * 8)];
+#pragma unroll
+        for (int mt = 0; mt < 4; ++mt) {
+          if (mt < mtiles) {
+            bf16x8 afrag = *(const bf16x8*)&la[swz64(mt * 16 + fr, ks * 32 + fk * 8)];
+            acc[sb][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[sb][mt], 0, 0, 0);
+          }
+        }
+      }
+    }
+    if (do_bias && db_f < F) {
+#pragma unroll
+      for (int kk = db_q * 16; kk < db_q * 16 + 16; ++kk) {
+        db_acc += __bfloat162float(__hip_bfloat16(__hip_bfloat16_raw{
+            (unsigned short)la[swz64(db_f, kk & ~7) + (kk & 7)]}));
+      }
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+#undef WG2_STAGE
+#undef WG2_ADVANCE
+
+  if (do_bias) {
+    __shared__ float db_lds[4][64];
+    db_lds[db_q][db_f] = db_acc;
+    __syncthreads();
+    if (db_q == 0 && db_f < F) {
+      const float v = db_lds[0][db_f] + db_lds[1][db_f] + db_lds[2][db_f] +
+                      db_lds[3][db_f];
+      if (DET) {
+        dBacc[((long)t * gridDim.y + blockIdx.y) * F + db_f] = v;
+      } else {
+        atomicAdd(&dBacc[(long)t * F + db_f], v);
+      }
+    }
+  }
+
+  float* dWt = DET ? dWacc + ((long)t * gridDim.y + blockIdx.y) * N9 * F
+                   : dWacc + (long)t * N9 * F;
+#pragma unroll
+  for (int sb = 0; sb < NSUB; ++sb)
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+    if (mt >= mtiles) continue;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int f = mt * 16 + fk * 4 + j;
+      const int n = n0 + (sb * 4 + wave) * 16 + fr;
+      if (f < F && n < N9) {
+        if (DET) {
+          dWt[(long)n * F + f] = acc[sb][mt][j];
+        } else {
+          atomicAdd(&dWt[(long)n * F + f], acc[sb][mt][j]);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // wgrad kernel: dWacc[t][n=9C][f] += sum_k dY[t,k,f] * im2col(X)[t,k,n]
 // Block: 256 thr = 4 waves; wave w owns n-subtile w (16 cols), iterates
 // m-tiles over F.  K-chunked grid with fp32 atomicAdd.
 //   dY [T, NB, Ho, Wo, F] bf16 ; X [T, NB, H, W, C] bf16
 // ---------------------------------------------------------------------------
-#define WG_KCHUNK 4096
-#define WBK 64   // wgrad K-step (2 MFMA K-slices per barrier)
-
-#define WGN 64   // wgrad output columns per block (4 waves x 16)
 
 // DET=true: each (t, K-chunk) block writes a PRIVATE accumulator slice
 // (plain stores, no atomics) — deterministic; DET=false: fp32 atomicAdd
@@ -940,6 +1250,104 @@ std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
                      T, NB, H, W, C, Ho, Wo, F, (int)pad, (int)kchunk)
   if (det) LAUNCH_WGRAD(true); else LAUNCH_WGRAD(false);
 #undef LAUNCH_WGRAD
+  auto dw = torch::empty({T, F, C, 3, 3}, x.options().dtype(torch::kFloat32));
+  const long total = (long)T * F * C * 9;
+  hipLaunchKernelGGL(wgrad_finalize_kernel, dim3(ew_grid2(total, 256)),
+                     dim3(256), 0, stream.stream(), acc.data_ptr<float>(),
+                     dw.data_ptr<float>(), T, F, C, nslices);
+  auto db = nslices == 1 ? dbacc.select(1, 0).contiguous() : dbacc.sum(1);
+  return {dw, db};
+}
+
+// wgrad v2: transpose both operands into k-contiguous layouts, then the
+// async-staged main kernel.  dy [T,NB,Ho,Wo,F], x [T,NB,H,W,C] ->
+// {dw [T,F,C,3,3] fp32, db [T,F] fp32}.  Requires Wo <= 64, F <= 64.
+std::vector<torch::Tensor> tconv_wgrad_v2(torch::Tensor dy, torch::Tensor x,
+                                          long pad, bool with_bias) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda());
+  auto dyc = dy.contiguous();
+  auto xc = x.contiguous();
+  const int T = (int)x.size(0), NB = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3), C = (int)x.size(4);
+  const int Ho = (int)dy.size(2), Wo = (int)dy.size(3), F = (int)dy.size(4);
+  TORCH_CHECK(F <= 64, "wgrad_v2 needs F <= 64");
+  const char* det_env = getenv("MAML355_DETERMINISTIC");
+  const bool det = det_env && det_env[0] == '1';
+  const int N9 = 9 * C;
+  const int Wo8 = ((Wo + 7) / 8) * 8;
+  const long Kprime = (long)NB * Ho * Wo8;
+  const long Kr = ((Kprime + WBK - 1) / WBK) * WBK;
+  const int Hxp = H + 2 * (int)pad;
+  const int Wxp = Wo8 + 2;
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  // operand transposes
+  auto dyt = torch::empty({T, F, Kr}, x.options());
+  {
+    const long blocks = (long)T * NB * Ho;
+    hipLaunchKernelGGL(dyt_pad_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(dyc.data_ptr()),
+                       reinterpret_cast<bf16*>(dyt.data_ptr()),
+                       T, NB, Ho, Wo, F, Wo8, Kr);
+    if (Kr > Kprime) {
+      const long total = (long)T * F * (Kr - Kprime);
+      hipLaunchKernelGGL(dyt_tail_kernel, dim3(ew_grid2(total, 256)), dim3(256),
+                         0, stream.stream(),
+                         reinterpret_cast<bf16*>(dyt.data_ptr()), T, F, Kprime,
+                         Kr);
+    }
+  }
+  auto xt = torch::empty({(long)T * C * NB * Hxp * Wxp}, x.options());
+  {
+    const long blocks = (long)T * NB * Hxp * ((Wxp + 63) / 64);
+    hipLaunchKernelGGL(xt_pad_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(xc.data_ptr()),
+                       reinterpret_cast<bf16*>(xt.data_ptr()),
+                       T, NB, H, W, C, Hxp, Wxp, (int)pad);
+  }
+  auto zpage = torch::zeros({16}, x.options());
+
+  // n-columns per block: NSUB=2 (128 cols, 2x MFMA per barrier) pays only
+  // on large-K shapes (measured: conv1-tgt 94->104 TF; SLOWER below
+  // Kr ~1.3e5 — grid starvation + 48 KB LDS occupancy drop).  Env
+  // MAML355_WGRAD_NSUB overrides for A/B.
+  int nsub = (N9 > 64 && Kr >= 131072) ? 2 : 1;
+  if (const char* e = getenv("MAML355_WGRAD_NSUB")) {
+    if (e[0] == '1') nsub = 1; else if (e[0] == '2' && N9 > 64) nsub = 2;
+  }
+  const int wgn = 64 * nsub;
+
+  // grid sizing as v1 (>= ~1024 blocks), K-chunks multiple of WBK
+  const int gridx = (N9 + wgn - 1) / wgn;
+  long desired_y = std::max<long>(1, 1024 / std::max<long>(1, (long)gridx * T));
+  long kchunk = (Kr + desired_y - 1) / desired_y;
+  kchunk = ((kchunk + WBK - 1) / WBK) * WBK;
+  kchunk = std::max<long>(kchunk, WBK);
+  if (!det) kchunk = std::min<long>(kchunk, (long)WG_KCHUNK);
+  const int gridy = (int)((Kr + kchunk - 1) / kchunk);
+  const int nslices = det ? gridy : 1;
+  auto acc = torch::zeros({T, nslices, N9, F},
+                          x.options().dtype(torch::kFloat32));
+  auto dbacc = torch::zeros({T, nslices, F},
+                            x.options().dtype(torch::kFloat32));
+  dim3 grid((unsigned)gridx, (unsigned)gridy, T);
+#define LAUNCH_WG2(NS_, DET_)                                                  \
+  hipLaunchKernelGGL((tconv_wgrad_v2_kernel<NS_, DET_>), grid, dim3(256), 0,   \
+                     stream.stream(),                                          \
+                     reinterpret_cast<const bf16*>(dyt.data_ptr()),            \
+                     reinterpret_cast<const bf16*>(xt.data_ptr()),             \
+                     reinterpret_cast<const bf16*>(zpage.data_ptr()),          \
+                     acc.data_ptr<float>(),                                    \
+                     with_bias ? dbacc.data_ptr<float>() : nullptr,            \
+                     T, NB, Ho, Wo8, Hxp, Wxp, C, F, Kr, Kprime, (int)kchunk)
+  if (nsub == 2) {
+    if (det) LAUNCH_WG2(2, true); else LAUNCH_WG2(2, false);
+  } else {
+    if (det) LAUNCH_WG2(1, true); else LAUNCH_WG2(1, false);
+  }
+#undef LAUNCH_WG2
   auto dw = torch::empty({T, F, C, 3, 3}, x.options().dtype(torch::kFloat32));
   const long total = (long)T * F * C * 9;
   hipLaunchKernelGGL(wgrad_finalize_kernel, dim3(ew_grid2(total, 256)),

@@ -136,3 +136,38 @@ def test_conv_v2_bitwise_matches_v1():
         dx2 = ext.tconv_mm_v2(dy, wpd2, None, 2 - pad, H, W, C, False)[0]
         assert (dx1 == dx2).all().item(), \
             f"v2 dgrad != v1 at {(T, NB, H, W, C, F, pad)}"
+
+
+def test_wgrad_v2_matches_v1():
+    """wgrad v2 (transposed operands + async staging) vs the v1 gather
+    kernel: same math, different fp32 accumulation order -> tight fp
+    tolerance, not bitwise.  Covers C in {1,3,48,64} incl. first-layer
+    shapes v1 staged scalar."""
+    from howtotrainyourmamlpytorch_amd.ops import hip_ext
+    ext = hip_ext()
+    torch.manual_seed(13)
+    for (T, NB, H, W, C, F, pad) in [(3, 7, 14, 14, 48, 48, 1),
+                                     (2, 9, 28, 28, 1, 48, 1),
+                                     (2, 5, 28, 28, 3, 48, 1),
+                                     (2, 6, 21, 21, 48, 48, 1),
+                                     (2, 4, 28, 28, 64, 64, 1),
+                                     (1, 3, 12, 12, 16, 32, 0)]:
+        Ho, Wo = H + 2 * pad - 2, W + 2 * pad - 2
+        x = torch.randn(T, NB, H, W, C, device="cuda", dtype=torch.bfloat16)
+        dy = torch.randn(T, NB, Ho, Wo, F, device="cuda", dtype=torch.bfloat16)
+        dw1, db1 = ext.tconv_wgrad(dy, x, pad, True)
+        dw2, db2 = ext.tconv_wgrad_v2(dy, x, pad, True)
+        torch.testing.assert_close(dw2, dw1, rtol=1e-4, atol=1e-2), (T, NB, C)
+        torch.testing.assert_close(db2, db1, rtol=1e-4, atol=1e-2)
+
+
+def test_wgrad_v2_deterministic(monkeypatch):
+    from howtotrainyourmamlpytorch_amd.ops import hip_ext
+    ext = hip_ext()
+    monkeypatch.setenv("MAML355_DETERMINISTIC", "1")
+    torch.manual_seed(14)
+    x = torch.randn(3, 9, 20, 20, 48, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(3, 9, 20, 20, 48, device="cuda", dtype=torch.bfloat16)
+    a = ext.tconv_wgrad_v2(dy, x, 1, True)
+    b = ext.tconv_wgrad_v2(dy, x, 1, True)
+    assert (a[0] == b[0]).all().item() and (a[1] == b[1]).all().item()

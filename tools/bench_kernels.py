@@ -54,6 +54,10 @@ def bench_conv(T, NB, H, W, C, F, tag):
         print(f"[conv {tag}]   v2:                 "
               f"fwd {t_f2*1e6:7.1f}us {flops/t_f2/1e12:6.1f}TF | "
               f"dgrad {t_d2*1e6:7.1f}us {flops/t_d2/1e12:6.1f}TF")
+    if F <= 64:
+        t_w2 = timeit(lambda: ext.tconv_wgrad_v2(dy, x, 1, True))
+        print(f"[conv {tag}]   wgrad_v2:           "
+              f"{t_w2*1e6:7.1f}us {flops/t_w2/1e12:6.1f}TF")
 
 
 def bench_bn(T, M, C, tag):
