@@ -409,6 +409,14 @@ def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
         bc = b.to(x.dtype) if b is not None else None
         y = ref.task_conv3x3(x, wc, bc, stride, padding)
         return (y, None) if return_stats else y
+    C = w.shape[2]
+    if C < 8:
+        # first-layer channels (1 or 3): zero-pad to 8 so the vectorized
+        # 8-channel staging paths apply (measured 25 TF on the scalar
+        # C=3 path).  F.pad is differentiable, so dw/dx slicing back to C
+        # channels is automatic (pad backward = narrow).
+        x = torch.nn.functional.pad(x, (0, 8 - C))
+        w = torch.nn.functional.pad(w, (0, 0, 0, 0, 0, 8 - C))
     y, sums = _ConvFwdFn.apply(x.contiguous(), w.contiguous(),
                                b.contiguous() if b is not None else None,
                                padding, return_stats)

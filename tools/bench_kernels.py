@@ -104,7 +104,28 @@ def main():
     bench_bn(8, 75 * 42 * 42, 48, "mi-bn1-tgt")
     bench_bn(8, 100 * 28 * 28, 64, "om-bn1")
     bench_pool(8 * 75, 84, 84, 48, "mi-pool0-tgt")
+    bench_conv0_wrapper(8, 75, 84, 84, 3, 48, "mi-conv0-tgt")
+    bench_conv0_wrapper(8, 100, 28, 28, 1, 64, "om-conv0-sup")
 
 
 if __name__ == "__main__":
     main()
+
+
+def bench_conv0_wrapper(T, NB, H, W, C, F, tag):
+    """First-layer conv through the dispatch wrapper (exercises the
+    channel-pad-to-8 path): fwd and fwd+wgrad."""
+    dev = torch.device("cuda")
+    x = torch.randn(T, NB, H, W, C, device=dev).to(torch.bfloat16)
+    w = torch.randn(T, F, C, 3, 3, device=dev, requires_grad=True)
+    b = torch.randn(T, F, device=dev, requires_grad=True)
+    flops = 2.0 * T * NB * H * W * F * 9 * C
+    t_f = timeit(lambda: ops.task_conv3x3(x, w, b), reps=20)
+
+    def fb():
+        y = ops.task_conv3x3(x, w, b)
+        torch.autograd.grad(y.float().sum(), [w, b])
+    t_fb = timeit(fb, reps=20)
+    print(f"[conv0w {tag}] T{T} NB{NB} {H}x{W} C{C}->F{F}  "
+          f"fwd {t_f*1e6:7.1f}us {flops/t_f/1e12:6.1f}TF | "
+          f"fwd+wgrad {t_fb*1e6:7.1f}us")
