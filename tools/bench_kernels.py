@@ -108,8 +108,6 @@ def main():
     bench_conv0_wrapper(8, 100, 28, 28, 1, 64, "om-conv0-sup")
 
 
-if __name__ == "__main__":
-    main()
 
 
 def bench_conv0_wrapper(T, NB, H, W, C, F, tag):
@@ -129,3 +127,7 @@ def bench_conv0_wrapper(T, NB, H, W, C, F, tag):
     print(f"[conv0w {tag}] T{T} NB{NB} {H}x{W} C{C}->F{F}  "
           f"fwd {t_f*1e6:7.1f}us {flops/t_f/1e12:6.1f}TF | "
           f"fwd+wgrad {t_fb*1e6:7.1f}us")
+
+
+if __name__ == "__main__":
+    main()
