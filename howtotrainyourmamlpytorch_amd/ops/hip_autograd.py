@@ -356,9 +356,9 @@ class _ConvWgradFn(torch.autograd.Function):
     def forward(ctx, dy, x, pad, with_bias):
         ctx.save_for_backward(dy, x)
         ctx.pad = pad
-        # v2: global transposes -> linear async staging (covers every C,
-        # incl. the first layer); v1 gather kernel for shapes v2 excludes
-        if (dy.shape[3] <= 64 and dy.shape[4] <= 64
+        # v2: global transposes -> linear async staging (covers every C
+        # and any Wo, incl. the 84x84 first layer); v1 for F > 64
+        if (dy.shape[4] <= 64
                 and os.environ.get("MAML355_WGRAD_V2", "1") != "0"):
             dw, db = _ext().tconv_wgrad_v2(dy, x, pad, with_bias)
         else:
