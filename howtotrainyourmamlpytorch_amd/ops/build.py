@@ -24,6 +24,7 @@ SOURCES = [
     os.path.join(HIP_DIR, "tconv.hip"),
     os.path.join(HIP_DIR, "bn_dbwd.hip"),
     os.path.join(HIP_DIR, "adam.hip"),
+    os.path.join(HIP_DIR, "dconv.hip"),
 ]
 
 

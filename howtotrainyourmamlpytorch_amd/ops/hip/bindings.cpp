@@ -46,6 +46,12 @@ std::vector<torch::Tensor> tconv_wgrad(torch::Tensor dy, torch::Tensor x,
 torch::Tensor tconv_repack_v2(torch::Tensor w, bool dgrad);
 std::vector<torch::Tensor> tconv_wgrad_v2(torch::Tensor dy, torch::Tensor x,
                                           long pad, bool with_bias);
+// dconv.hip
+torch::Tensor dconv_fwd(torch::Tensor x, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias, long pad,
+                        long Ho, long Wo);
+std::vector<torch::Tensor> dconv_wgrad(torch::Tensor dy, torch::Tensor x,
+                                       long pad, bool with_bias);
 std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
                                        c10::optional<torch::Tensor> bias,
                                        long pad, long Ho, long Wo, long Co,
@@ -81,6 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "async-pipelined MFMA 3x3 conv fwd/dgrad (v2)");
   m.def("tconv_wgrad_v2", &tconv_wgrad_v2,
         "async-pipelined wgrad via k-contiguous operand transposes (v2)");
+  m.def("dconv_fwd", &dconv_fwd, "direct small-C conv fwd (VALU)");
+  m.def("dconv_wgrad", &dconv_wgrad, "direct small-C conv wgrad (VALU)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam + grad clamp");
 }

@@ -268,6 +268,13 @@ def lslr_update(arena, grad, lr_vec):
 # custom kernels at EVERY derivative order (second-order MAML's
 # create_graph included), no torch fallback on the hot path.
 # ---------------------------------------------------------------------------
+def _dconv_ok(ci: int, f: int) -> bool:
+    """Direct VALU conv for small input-channel counts (the first layer's
+    C in {1,3}); MAML355_NO_DCONV=1 reverts to the GEMM path for A/B."""
+    return (ci <= 8 and f in (16, 32, 48, 64)
+            and os.environ.get("MAML355_NO_DCONV", "0") != "1")
+
+
 def _conv_v2_ok(ci: int) -> bool:
     """v2 (async global_load_lds pipeline) needs 8-aligned input channels.
     Measured SLOWER than v1 at the flagship shapes (156 vs 193 TF conv1:
@@ -287,7 +294,12 @@ class _ConvFwdFn(torch.autograd.Function):
         ctx.has_bias = b is not None
         H, W = x.shape[2], x.shape[3]
         Ho, Wo = H + 2 * pad - 2, W + 2 * pad - 2
-        if _conv_v2_ok(x.shape[4]):
+        if _dconv_ok(x.shape[4], w.shape[1]):
+            # first-layer shapes: direct VALU conv (im2col-MFMA wastes
+            # >4x the MACs at K = 9..27)
+            y = _ext().dconv_fwd(x, w, b, pad, Ho, Wo)
+            sums = torch.empty(0, device=x.device, dtype=torch.float32)
+        elif _conv_v2_ok(x.shape[4]):
             wp = _ext().tconv_repack_v2(w, False)
             y, sums = _ext().tconv_mm_v2(x, wp, b, pad, Ho, Wo, w.shape[1],
                                          want_stats)
@@ -356,9 +368,12 @@ class _ConvWgradFn(torch.autograd.Function):
     def forward(ctx, dy, x, pad, with_bias):
         ctx.save_for_backward(dy, x)
         ctx.pad = pad
-        # v2: global transposes -> linear async staging (covers every C
-        # and any Wo, incl. the 84x84 first layer); v1 for F > 64
-        if (dy.shape[4] <= 64
+        # small-C shapes: direct VALU wgrad; else v2 (global transposes ->
+        # linear async staging, any Wo); v1 for F > 64
+        if (x.shape[4] <= 8 and dy.shape[4] <= 64
+                and os.environ.get("MAML355_NO_DCONV", "0") != "1"):
+            dw, db = _ext().dconv_wgrad(dy, x, pad, with_bias)
+        elif (dy.shape[4] <= 64
                 and os.environ.get("MAML355_WGRAD_V2", "1") != "0"):
             dw, db = _ext().tconv_wgrad_v2(dy, x, pad, with_bias)
         else:
@@ -409,12 +424,11 @@ def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
         bc = b.to(x.dtype) if b is not None else None
         y = ref.task_conv3x3(x, wc, bc, stride, padding)
         return (y, None) if return_stats else y
-    C = w.shape[2]
-    if C < 8:
-        # first-layer channels (1 or 3): zero-pad to 8 so the vectorized
-        # 8-channel staging paths apply (measured 25 TF on the scalar
-        # C=3 path).  F.pad is differentiable, so dw/dx slicing back to C
-        # channels is automatic (pad backward = narrow).
+    C, F = w.shape[2], w.shape[1]
+    if C < 8 and not _dconv_ok(C, F):
+        # small C without a direct-kernel instantiation: zero-pad to 8 so
+        # the vectorized 8-channel GEMM staging applies.  F.pad is
+        # differentiable, so dw/dx slicing back to C channels is automatic.
         x = torch.nn.functional.pad(x, (0, 8 - C))
         w = torch.nn.functional.pad(w, (0, 0, 0, 0, 0, 8 - C))
     y, sums = _ConvFwdFn.apply(x.contiguous(), w.contiguous(),

@@ -171,3 +171,39 @@ def test_wgrad_v2_deterministic(monkeypatch):
     a = ext.tconv_wgrad_v2(dy, x, 1, True)
     b = ext.tconv_wgrad_v2(dy, x, 1, True)
     assert (a[0] == b[0]).all().item() and (a[1] == b[1]).all().item()
+
+
+def test_dconv_matches_gemm_path():
+    """Direct small-C kernels vs the GEMM reference: fwd tight-bf16 equal,
+    wgrad to fp accumulation tolerance."""
+    from howtotrainyourmamlpytorch_amd.ops import hip_ext, reference as ref
+    ext = hip_ext()
+    torch.manual_seed(17)
+    for (T, NB, H, W, C, F, pad) in [(2, 5, 28, 28, 1, 64, 1),
+                                     (3, 7, 28, 28, 3, 48, 1),
+                                     (2, 4, 12, 12, 3, 16, 1),
+                                     (2, 3, 10, 10, 8, 32, 0)]:
+        Ho, Wo = H + 2 * pad - 2, W + 2 * pad - 2
+        x = torch.randn(T, NB, H, W, C, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(T, F, C, 3, 3, device="cuda")
+        b = torch.randn(T, F, device="cuda")
+        y = ext.dconv_fwd(x, w, b, pad, Ho, Wo)
+        yr = ref.task_conv3x3(x.float().cpu(), w.cpu(), b.cpu(), 1, pad)
+        torch.testing.assert_close(y.float().cpu(), yr, rtol=2e-2, atol=2e-2)
+        dy = torch.randn(T, NB, Ho, Wo, F, device="cuda", dtype=torch.bfloat16)
+        dw, db = ext.dconv_wgrad(dy, x, pad, True)
+        dw1, db1 = ext.tconv_wgrad(dy, x, pad, True)
+        torch.testing.assert_close(dw, dw1, rtol=1e-4, atol=1e-2)
+        torch.testing.assert_close(db, db1, rtol=1e-4, atol=1e-2)
+
+
+def test_dconv_wgrad_deterministic(monkeypatch):
+    from howtotrainyourmamlpytorch_amd.ops import hip_ext
+    ext = hip_ext()
+    monkeypatch.setenv("MAML355_DETERMINISTIC", "1")
+    torch.manual_seed(18)
+    x = torch.randn(3, 9, 28, 28, 3, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(3, 9, 28, 28, 48, device="cuda", dtype=torch.bfloat16)
+    a = ext.dconv_wgrad(dy, x, 1, True)
+    b = ext.dconv_wgrad(dy, x, 1, True)
+    assert (a[0] == b[0]).all().item() and (a[1] == b[1]).all().item()
