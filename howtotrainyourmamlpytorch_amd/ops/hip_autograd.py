@@ -368,10 +368,12 @@ class _ConvWgradFn(torch.autograd.Function):
     def forward(ctx, dy, x, pad, with_bias):
         ctx.save_for_backward(dy, x)
         ctx.pad = pad
-        # small-C shapes: direct VALU wgrad; else v2 (global transposes ->
-        # linear async staging, any Wo); v1 for F > 64
+        # v2 (global transposes -> linear async staging, any Wo incl. the
+        # first layer); v1 for F > 64.  The direct VALU wgrad
+        # (dconv_wgrad) measured 2.3x SLOWER than v2 at the conv0 shape
+        # (serial position walk per wave) — opt-in only.
         if (x.shape[4] <= 8 and dy.shape[4] <= 64
-                and os.environ.get("MAML355_NO_DCONV", "0") != "1"):
+                and os.environ.get("MAML355_DCONV_WGRAD", "0") == "1"):
             dw, db = _ext().dconv_wgrad(dy, x, pad, with_bias)
         elif (dy.shape[4] <= 64
                 and os.environ.get("MAML355_WGRAD_V2", "1") != "0"):
