@@ -39,6 +39,12 @@ def maybe_unzip_dataset(args) -> None:
     """Ensure ``args.dataset_path`` exists and contains the expected number
     of files; otherwise (re-)extract ``<dataset_path>.tar.bz2``."""
     path = args.dataset_path
+    # compact-npz backend needs no extraction
+    if str(path).endswith(".npz") and os.path.isfile(path):
+        return
+    if getattr(args, "load_from_npz_files", False) and \
+            os.path.isfile(str(path).rstrip(os.sep) + ".npz"):
+        return
     expected = EXPECTED_FILE_COUNTS.get(args.dataset_name)
     ok = os.path.isdir(path) and (expected is None or count_files(path) >= expected)
     if ok:
