@@ -46,22 +46,28 @@ __global__ void bn_sums_vec_kernel(const scalar_t* __restrict__ x,
     }
   }
   __syncthreads();
-  // in-block reduce in FIXED row-group order (bitwise deterministic)
-  if (rg == 0) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) { s[j] = 0.f; q[j] = 0.f; }
-    for (int rr = 0; rr < rows_in_block; ++rr) {
+  // in-block pairwise tree reduce with FIXED pairing order (bitwise
+  // deterministic, log2(rows) steps instead of a serial row loop)
+  for (int n_ = rows_in_block; n_ > 1;) {
+    const int step = (n_ + 1) >> 1;   // non-pow2-safe fixed pairing
+    if (rg < step && rg + step < n_) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        s[j] += ls[(rr * 2 + 0) * C + c8 * 8 + j];
-        q[j] += ls[(rr * 2 + 1) * C + c8 * 8 + j];
+        ls[(rg * 2 + 0) * C + c8 * 8 + j] +=
+            ls[((rg + step) * 2 + 0) * C + c8 * 8 + j];
+        ls[(rg * 2 + 1) * C + c8 * 8 + j] +=
+            ls[((rg + step) * 2 + 1) * C + c8 * 8 + j];
       }
     }
+    __syncthreads();
+    n_ = step;
+  }
+  if (rg == 0) {
     float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 2) * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      pt[c8 * 8 + j] = s[j];
-      pt[C + c8 * 8 + j] = q[j];
+      pt[c8 * 8 + j] = ls[0 * C + c8 * 8 + j];
+      pt[C + c8 * 8 + j] = ls[1 * C + c8 * 8 + j];
     }
   }
 }
@@ -239,21 +245,26 @@ __global__ void bn_bwd_sums_vec_kernel(const scalar_t* __restrict__ dy,
     }
   }
   __syncthreads();
-  if (rg == 0) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) { s1[j] = 0.f; s2[j] = 0.f; }
-    for (int rr = 0; rr < rows_in_block; ++rr) {
+  for (int n_ = rows_in_block; n_ > 1;) {
+    const int step = (n_ + 1) >> 1;   // non-pow2-safe fixed pairing
+    if (rg < step && rg + step < n_) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        s1[j] += ls[(rr * 2 + 0) * C + c8 * 8 + j];
-        s2[j] += ls[(rr * 2 + 1) * C + c8 * 8 + j];
+        ls[(rg * 2 + 0) * C + c8 * 8 + j] +=
+            ls[((rg + step) * 2 + 0) * C + c8 * 8 + j];
+        ls[(rg * 2 + 1) * C + c8 * 8 + j] +=
+            ls[((rg + step) * 2 + 1) * C + c8 * 8 + j];
       }
     }
+    __syncthreads();
+    n_ = step;
+  }
+  if (rg == 0) {
     float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 2) * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      pt[c8 * 8 + j] = s1[j];
-      pt[C + c8 * 8 + j] = s2[j];
+      pt[c8 * 8 + j] = ls[0 * C + c8 * 8 + j];
+      pt[C + c8 * 8 + j] = ls[1 * C + c8 * 8 + j];
     }
   }
 }
@@ -484,21 +495,26 @@ __global__ void bn_pool_bwd_sums_vec_kernel(
     }
   }
   __syncthreads();
-  if (rg == 0) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) { s1[j] = 0.f; s2[j] = 0.f; }
-    for (int rr = 0; rr < rows_in_block; ++rr) {
+  for (int n_ = rows_in_block; n_ > 1;) {
+    const int step = (n_ + 1) >> 1;   // non-pow2-safe fixed pairing
+    if (rg < step && rg + step < n_) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        s1[j] += ls[(rr * 2 + 0) * C + c0 + j];
-        s2[j] += ls[(rr * 2 + 1) * C + c0 + j];
+        ls[(rg * 2 + 0) * C + c0 + j] +=
+            ls[((rg + step) * 2 + 0) * C + c0 + j];
+        ls[(rg * 2 + 1) * C + c0 + j] +=
+            ls[((rg + step) * 2 + 1) * C + c0 + j];
       }
     }
+    __syncthreads();
+    n_ = step;
+  }
+  if (rg == 0) {
     float* pt = partials + (((long)t * gridDim.y + blockIdx.y) * 2) * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      pt[c0 + j] = s1[j];
-      pt[C + c0 + j] = s2[j];
+      pt[c0 + j] = ls[0 * C + c0 + j];
+      pt[C + c0 + j] = ls[1 * C + c0 + j];
     }
   }
 }

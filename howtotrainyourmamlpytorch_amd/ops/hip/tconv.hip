@@ -334,7 +334,13 @@ void tconv_mm_kernel(const bf16* __restrict__ X, const bf16* __restrict__ Wp,
 // LDS: 2x(256x64) A + 2x(64x64) B bf16 = 80 KB dynamic -> 2 blocks/CU.
 // ---------------------------------------------------------------------------
 #define V2_LDS_BYTES (2 * BM * BK * 2 + 2 * 64 * BK * 2)
+#define V2_LDS_BYTES_SBUF (BM * BK * 2 + 64 * BK * 2)
 
+// DBUF=true: double-buffered 2-phase (stage ks+1 while computing ks,
+// 80 KB LDS -> 2 blocks/CU); DBUF=false: single buffer, stage->compute
+// per step (40 KB -> 3-4 blocks/CU; the async DMA still replaces the
+// VALU register staging).
+template <bool DBUF>
 __global__ __launch_bounds__(512, 2)
 void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wimg,
                         const float* __restrict__ bias, const bf16* __restrict__ zpage,
@@ -349,9 +355,10 @@ void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wi
   const int ntiles = (Co + 15) / 16;
 
   extern __shared__ short smem[];
-  // layout: A buffers at [0, BM*BK), [BM*BK, 2*BM*BK); B buffers after
-#define LDS_A(buf_) (smem + (buf_) * (BM * BK))
-#define LDS_B(buf_) (smem + 2 * (BM * BK) + (buf_) * (64 * BK))
+  // layout: A buffer(s) first, then B buffer(s)
+#define LDS_A(buf_) (smem + (DBUF ? (buf_) * (BM * BK) : 0))
+#define LDS_B(buf_) \
+  (smem + (DBUF ? 2 : 1) * (BM * BK) + (DBUF ? (buf_) * (64 * BK) : 0))
 
   const bf16* Xt = Xp + (long)t * NB * Hp * Wp * Ci;
   const short* Wt = (const short*)Wimg + (long)t * ksteps * 64 * BK;
@@ -363,14 +370,22 @@ void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wi
 
   // loop-invariant per-thread A-staging descriptors: 4 slots of 16B; slot
   // s -> (row m = s>>3, dest 8-col group j = s&7); source col group is
-  // j ^ (m&7) (inverse swizzle on the source side)
+  // j ^ (m&7) (inverse swizzle on the source side).  The k -> (dy,dx,c)
+  // decomposition advances INCREMENTALLY per K-step (adds + compares,
+  // no division in the loop).
   int a_rb[4];   // element offset of (n, ho, wo) in the padded input, or -1
-  int a_k8[4];   // source k base = (j ^ (m&7)) * 8 within the K-step
+  int a_c[4], a_dy[4], a_dx[4], a_koff[4], a_k[4];
 #pragma unroll
   for (int q = 0; q < 4; ++q) {
     const int s = q * 512 + threadIdx.x;
     const int m = s >> 3;
-    a_k8[q] = ((s & 7) ^ (m & 7)) * 8;
+    const int k8 = ((s & 7) ^ (m & 7)) * 8;   // k base within the K-step
+    a_k[q] = k8;
+    const int kyx = k8 / Ci;
+    a_c[q] = k8 - kyx * Ci;
+    a_dy[q] = kyx / 3;
+    a_dx[q] = kyx % 3;
+    a_koff[q] = (a_dy[q] * Wp + a_dx[q]) * Ci + a_c[q];
     const long mg = m0 + m;
     if (mg < Mtot) {
       const int wo = (int)(mg % Wo);
@@ -381,21 +396,27 @@ void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wi
       a_rb[q] = -1;
     }
   }
+#define V2_ADVANCE()                                                           \
+  do {                                                                         \
+    _Pragma("unroll")                                                          \
+    for (int q = 0; q < 4; ++q) {                                              \
+      a_k[q] += BK;                                                            \
+      a_c[q] += BK;                                                            \
+      while (a_c[q] >= Ci) {                                                   \
+        a_c[q] -= Ci;                                                          \
+        if (++a_dx[q] == 3) { a_dx[q] = 0; ++a_dy[q]; }                        \
+      }                                                                        \
+      a_koff[q] = (a_dy[q] * Wp + a_dx[q]) * Ci + a_c[q];                      \
+    }                                                                          \
+  } while (0)
 
 #define V2_STAGE(ks_, buf_)                                                    \
   do {                                                                         \
-    const int k0_ = (ks_)*BK;                                                  \
     _Pragma("unroll")                                                          \
     for (int q = 0; q < 4; ++q) {                                              \
-      const int k = k0_ + a_k8[q];                                             \
-      const bf16* src;                                                         \
-      if (a_rb[q] < 0 || k >= K9) {                                            \
-        src = zpage;                                                           \
-      } else {                                                                 \
-        const int kyx = k / Ci;                                                \
-        const int c = k - kyx * Ci;                                            \
-        src = Xt + a_rb[q] + ((kyx / 3) * Wp + (kyx % 3)) * Ci + c;            \
-      }                                                                        \
+      const bf16* src = (a_rb[q] < 0 || a_k[q] >= K9)                          \
+                            ? zpage                                            \
+                            : Xt + a_rb[q] + a_koff[q];                        \
       auto ldst = (__attribute__((address_space(3))) void*)(                   \
           &LDS_A(buf_)[((long)q * 512 + wave * 64) * 8]);                      \
       __builtin_amdgcn_global_load_lds(                                        \
@@ -416,12 +437,20 @@ void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wi
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[hh][i] = {0.f, 0.f, 0.f, 0.f};
 
-  V2_STAGE(0, 0);
-  __syncthreads();
-
   int buf = 0;
+  if (DBUF) {
+    V2_STAGE(0, 0);
+    __syncthreads();
+  }
   for (int ks = 0; ks < ksteps; ++ks) {
-    if (ks + 1 < ksteps) V2_STAGE(ks + 1, buf ^ 1);
+    if (DBUF) {
+      V2_ADVANCE();
+      if (ks + 1 < ksteps) V2_STAGE(ks + 1, buf ^ 1);
+    } else {
+      V2_STAGE(ks, 0);
+      V2_ADVANCE();
+      __syncthreads();
+    }
     const short* la = LDS_A(buf);
     const short* lb = LDS_B(buf);
 #pragma unroll
@@ -437,10 +466,11 @@ void tconv_mm_v2_kernel(const bf16* __restrict__ Xp, const bf16* __restrict__ Wi
         }
       }
     }
-    __syncthreads();  // drains vmcnt -> buf^1 staged; all reads of buf done
-    buf ^= 1;
+    __syncthreads();  // drains vmcnt; all reads of buf done
+    if (DBUF) buf ^= 1;
   }
 #undef V2_STAGE
+#undef V2_ADVANCE
 
   // epilogue — identical to v1
   __shared__ float sums_lds[2][64];
@@ -1187,21 +1217,30 @@ std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
   }
   const long Mtot = (long)NB * Ho * Wo;
   dim3 grid((unsigned)((Mtot + BM - 1) / BM), T);
+  const char* sbuf_env = getenv("MAML355_CONV_V2_SBUF");
+  const bool sbuf = sbuf_env && sbuf_env[0] == '1';
   static bool attr_set = false;
   if (!attr_set) {
-    hipFuncSetAttribute((const void*)tconv_mm_v2_kernel,
+    hipFuncSetAttribute((const void*)tconv_mm_v2_kernel<true>,
                         hipFuncAttributeMaxDynamicSharedMemorySize,
                         V2_LDS_BYTES);
+    hipFuncSetAttribute((const void*)tconv_mm_v2_kernel<false>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        V2_LDS_BYTES_SBUF);
     attr_set = true;
   }
-  hipLaunchKernelGGL(tconv_mm_v2_kernel, grid, dim3(512), V2_LDS_BYTES,
-                     stream.stream(),
-                     reinterpret_cast<const bf16*>(xp.data_ptr()),
-                     reinterpret_cast<const bf16*>(wimg.data_ptr()), bptr,
-                     reinterpret_cast<const bf16*>(zpage_cache.data_ptr()),
-                     reinterpret_cast<bf16*>(y.data_ptr()),
-                     with_stats ? sums.data_ptr<float>() : nullptr,
-                     T, NB, Hp, Wp, Ci, (int)Ho, (int)Wo, (int)Co);
+#define LAUNCH_MMV2(DB_, BYTES_)                                               \
+  hipLaunchKernelGGL((tconv_mm_v2_kernel<DB_>), grid, dim3(512), BYTES_,       \
+                     stream.stream(),                                          \
+                     reinterpret_cast<const bf16*>(xp.data_ptr()),             \
+                     reinterpret_cast<const bf16*>(wimg.data_ptr()), bptr,     \
+                     reinterpret_cast<const bf16*>(zpage_cache.data_ptr()),    \
+                     reinterpret_cast<bf16*>(y.data_ptr()),                    \
+                     with_stats ? sums.data_ptr<float>() : nullptr,            \
+                     T, NB, Hp, Wp, Ci, (int)Ho, (int)Wo, (int)Co)
+  if (sbuf) LAUNCH_MMV2(false, V2_LDS_BYTES_SBUF);
+  else LAUNCH_MMV2(true, V2_LDS_BYTES);
+#undef LAUNCH_MMV2
   return {y, sums};
 }
 
