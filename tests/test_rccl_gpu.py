@@ -74,3 +74,36 @@ def test_rccl_two_ranks(tmp_path):
     for p in procs:
         assert p.exitcode == 0, f"rank failed with {p.exitcode}"
     assert (tmp_path / "rccl_ok.pt").is_file()
+
+
+def test_rccl_single_rank_init_and_collectives(tmp_path):
+    """1-rank RCCL process group on one MI355X: exercises RCCL comm
+    creation and the DistContext collective code paths on hardware even
+    when no second GPU exists (the 2-rank test above covers real
+    communication when the driver has a multi-GPU node)."""
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", init_method=f"file://{tmp_path}/rdv1",
+                            rank=0, world_size=1)
+    try:
+        from howtotrainyourmamlpytorch_amd.parallel.dist import DistContext
+        ctx = DistContext(0, 1, 0, "nccl")
+        p = torch.nn.Parameter(torch.randn(300, device="cuda"))
+        p.grad = torch.randn_like(p)
+        g0 = p.grad.clone()
+        # world_size=1 short-circuits the flat bucket; call the raw
+        # collective directly so RCCL actually executes
+        t = p.grad.clone()
+        dist.all_reduce(t)
+        torch.testing.assert_close(t, g0)
+        n, s, q = ctx.all_reduce_sum_vector([1.0, 2.0, 3.0])
+        assert (n, s, q) == (1.0, 2.0, 3.0)
+        ctx.start_overlapped_reduction([p])
+        ctx.reduce_chunk_gradients([p])
+        ctx.finish_overlapped_reduction([p])
+        torch.testing.assert_close(p.grad, g0)
+        ctx.barrier()
+    finally:
+        dist.destroy_process_group()
