@@ -1217,8 +1217,10 @@ std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
   }
   const long Mtot = (long)NB * Ho * Wo;
   dim3 grid((unsigned)((Mtot + BM - 1) / BM), T);
+  // single-buffer is the measured default (conv1 192->222 TF vs v1;
+  // the 80 KB double-buffer drops occupancy 3->2 blocks/CU and loses)
   const char* sbuf_env = getenv("MAML355_CONV_V2_SBUF");
-  const bool sbuf = sbuf_env && sbuf_env[0] == '1';
+  const bool sbuf = !(sbuf_env && sbuf_env[0] == '0');
   static bool attr_set = false;
   if (!attr_set) {
     hipFuncSetAttribute((const void*)tconv_mm_v2_kernel<true>,

@@ -276,11 +276,12 @@ def _dconv_ok(ci: int, f: int) -> bool:
 
 
 def _conv_v2_ok(ci: int) -> bool:
-    """v2 (async global_load_lds pipeline) needs 8-aligned input channels.
-    Measured SLOWER than v1 at the flagship shapes (156 vs 193 TF conv1:
-    80 KB LDS drops occupancy 3->2 blocks/CU and the in-loop k-offset
-    divisions add VALU) — opt-in via MAML355_CONV_V2=1 until it wins."""
-    return ci % 8 == 0 and os.environ.get("MAML355_CONV_V2", "0") == "1"
+    """v2 (async global_load_lds staging, single-buffer default) needs
+    8-aligned input channels.  Measured: SBUF v2 beats v1 at every
+    flagship shape (conv1 192->222 TF, conv2 146->175, omniglot
+    289->312); the 80 KB double-buffer variant (MAML355_CONV_V2_SBUF=0)
+    loses to occupancy.  MAML355_CONV_V2=0 reverts to v1."""
+    return ci % 8 == 0 and os.environ.get("MAML355_CONV_V2", "1") != "0"
 
 
 class _ConvFwdFn(torch.autograd.Function):

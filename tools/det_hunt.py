@@ -107,5 +107,65 @@ def main():
     hunt("meta_grads_x30", meta, 30)
 
 
+def hunt_trainloop_and_fresh_linear():
+    """Targets the Adam-amplified full-train-loop nondeterminism: (a) the
+    exact failing-test scenario repeated; (b) linear bmm with FRESH
+    allocations per trial (allocator-address-dependent algorithm choice
+    would escape same-tensor trials)."""
+    from howtotrainyourmamlpytorch_amd.ops import reference as ref
+
+    def train_once():
+        torch.manual_seed(123)
+        args = get_args([
+            "--batch_size", "4", "--num_classes_per_set", "5",
+            "--num_samples_per_class", "1", "--num_target_samples", "3",
+            "--image_height", "28", "--image_width", "28", "--image_channels", "1",
+            "--cnn_num_filters", "48",
+            "--number_of_training_steps_per_iter", "3",
+            "--second_order", "True", "--first_order_to_second_order_epoch", "-1",
+            "--total_epochs", "5", "--seed", "7", "--dataset_name", "synthetic",
+        ])
+        model = MAMLFewShotClassifier(im_shape=(2, 1, 28, 28), device=dev, args=args)
+        stream = SyntheticEpisodeStream(args)
+        for batch in stream.get_train_batches(3):
+            model.run_train_iter(batch, epoch=0)
+        torch.cuda.synchronize()
+        return model.classifier.theta.detach().clone()
+
+    ref_theta = train_once()
+    bad = 0
+    for i in range(6):
+        t = train_once()
+        if not (t == ref_theta).all().item():
+            bad += 1
+            print(f"  trainloop trial {i}: max|d|={(t-ref_theta).abs().max().item():e}",
+                  flush=True)
+    print(f"[trainloop x6] {bad}/6 mismatching", flush=True)
+
+    cpu_lw = torch.randn(4, 5, 48)
+    cpu_f = torch.randn(4, 15, 48)
+    ref_out = None
+    bad = 0
+    for i in range(50):
+        # fresh GPU allocations with allocator perturbation
+        junk = torch.randn((i % 7 + 1) * 333, device=dev)  # noqa: F841
+        lw = cpu_lw.to(dev)
+        f = cpu_f.to(dev).to(torch.bfloat16)
+        lw2 = lw.requires_grad_()
+        f2 = f.requires_grad_()
+        out = ref.task_linear(f2, lw2.to(f.dtype), None)
+        out.float().square().sum().backward()
+        res = [out.detach().cpu(), lw2.grad.cpu(), f2.grad.cpu()]
+        if ref_out is None:
+            ref_out = res
+        elif any(not (a == b).all().item() for a, b in zip(res, ref_out)):
+            bad += 1
+    print(f"[linear_freshalloc x50] {bad}/50 mismatching", flush=True)
+
+
 if __name__ == "__main__":
-    main()
+    if len(sys.argv) > 1 and sys.argv[1] == "trainloop":
+        hunt_trainloop_and_fresh_linear()
+    else:
+        main()
+        hunt_trainloop_and_fresh_linear()
