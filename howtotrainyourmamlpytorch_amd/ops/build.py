@@ -25,6 +25,7 @@ SOURCES = [
     os.path.join(HIP_DIR, "bn_dbwd.hip"),
     os.path.join(HIP_DIR, "adam.hip"),
     os.path.join(HIP_DIR, "dconv.hip"),
+    os.path.join(HIP_DIR, "linear.hip"),
 ]
 
 

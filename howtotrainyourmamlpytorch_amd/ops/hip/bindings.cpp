@@ -52,6 +52,12 @@ torch::Tensor dconv_fwd(torch::Tensor x, torch::Tensor w,
                         long Ho, long Wo);
 std::vector<torch::Tensor> dconv_wgrad(torch::Tensor dy, torch::Tensor x,
                                        long pad, bool with_bias);
+// linear.hip
+torch::Tensor lin_fwd(torch::Tensor x, torch::Tensor w,
+                      c10::optional<torch::Tensor> b);
+torch::Tensor lin_dx(torch::Tensor dy, torch::Tensor w);
+std::vector<torch::Tensor> lin_wgrad(torch::Tensor dy, torch::Tensor x,
+                                     bool with_bias);
 std::vector<torch::Tensor> tconv_mm_v2(torch::Tensor x, torch::Tensor wimg,
                                        c10::optional<torch::Tensor> bias,
                                        long pad, long Ho, long Wo, long Co,
@@ -89,6 +95,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "async-pipelined wgrad via k-contiguous operand transposes (v2)");
   m.def("dconv_fwd", &dconv_fwd, "direct small-C conv fwd (VALU)");
   m.def("dconv_wgrad", &dconv_wgrad, "direct small-C conv wgrad (VALU)");
+  m.def("lin_fwd", &lin_fwd, "task-batched linear head fwd");
+  m.def("lin_dx", &lin_dx, "linear head input-grad");
+  m.def("lin_wgrad", &lin_wgrad, "linear head weight/bias grads");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam + grad clamp");
 }

@@ -440,8 +440,85 @@ def task_conv3x3(x, w, b=None, stride=1, padding=1, return_stats=False):
     return (y, sums) if return_stats else y
 
 
+# ---------------------------------------------------------------------------
+# linear head trio — the ops are mutually bilinear (like the convs), so
+# each backward composes the other two: custom kernels at every order.
+# Replaces torch.bmm (hipBLASLt autotunes on first call: run #1 of a
+# process differed from runs #2+ — measured 5e-3 theta drift).
+# ---------------------------------------------------------------------------
+class _LinFwdFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_b = b is not None
+        return _ext().lin_fwd(x, w, b)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = _LinDxFn.apply(dy, w)
+        want_db = ctx.has_b and ctx.needs_input_grad[2]
+        if ctx.needs_input_grad[1] or want_db:
+            dw_, db_ = _LinWgradFn.apply(dy, x, want_db)
+            if ctx.needs_input_grad[1]:
+                dw = dw_
+            if want_db:
+                db = db_
+        return dx, dw, db
+
+
+class _LinDxFn(torch.autograd.Function):
+    """dx = dy @ bf16(w) — bilinear in (dy, w)."""
+
+    @staticmethod
+    def forward(ctx, dy, w):
+        ctx.save_for_backward(dy, w)
+        return _ext().lin_dx(dy, w)
+
+    @staticmethod
+    def backward(ctx, g):
+        dy, w = ctx.saved_tensors
+        g = g.contiguous()
+        d_dy = d_w = None
+        if ctx.needs_input_grad[0]:
+            d_dy = _LinFwdFn.apply(g, w, None)
+        if ctx.needs_input_grad[1]:
+            d_w = _LinWgradFn.apply(dy, g, False)[0]
+        return d_dy, d_w
+
+
+class _LinWgradFn(torch.autograd.Function):
+    """(dy, x) -> (dw = dy^T @ x, db = sum dy) — bilinear in (dy, x)."""
+
+    @staticmethod
+    def forward(ctx, dy, x, with_bias):
+        ctx.save_for_backward(dy, x)
+        dw, db = _ext().lin_wgrad(dy, x, with_bias)
+        return dw, db
+
+    @staticmethod
+    def backward(ctx, gw, gdb):
+        dy, x = ctx.saved_tensors
+        d_dy = d_x = None
+        if ctx.needs_input_grad[0]:
+            if gw is not None:
+                d_dy = _LinFwdFn.apply(x, gw.contiguous(), None)
+            if gdb is not None:
+                add = gdb.unsqueeze(1).to(dy.dtype)
+                d_dy = add.expand_as(dy).contiguous() if d_dy is None else d_dy + add
+        if ctx.needs_input_grad[1] and gw is not None:
+            d_x = _LinDxFn.apply(dy, gw.contiguous())
+        return d_dy, d_x, None
+
+
 def task_linear(x, w, b=None):
-    return ref.task_linear(x, w.to(x.dtype), b.to(x.dtype) if b is not None else None)
+    if x.dtype != torch.bfloat16:
+        # fp32 oracle mode on GPU: torch composition
+        return ref.task_linear(x, w.to(x.dtype), b.to(x.dtype) if b is not None else None)
+    return _LinFwdFn.apply(x.contiguous(), w, b)
 
 
 def fused_adam_step(params, grads, exp_avgs, exp_avg_sqs, step, lr,

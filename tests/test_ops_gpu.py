@@ -343,3 +343,42 @@ def test_fused_adam_optimizer_runs_on_gpu():
         torch.testing.assert_close(a.detach().cpu(), b.detach(), rtol=1e-4, atol=1e-5)
     sd = opt_g.state_dict()
     assert "exp_avg" in list(sd["state"].values())[0]
+
+
+def test_linear_trio_matches_reference():
+    """lin_fwd/dx/wgrad vs the torch oracle incl. second-order compose."""
+    torch.manual_seed(21)
+    T, M, K, ways = 3, 17, 200, 5
+    x = torch.randn(T, M, K, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(T, ways, K, device=dev(), requires_grad=True)
+    b = torch.randn(T, ways, device=dev(), requires_grad=True)
+    y = ops.task_linear(x, w, b)
+    xr = x.detach().float().cpu().requires_grad_()
+    wr = w.detach().cpu().requires_grad_()
+    br = b.detach().cpu().requires_grad_()
+    yr = ref.task_linear(xr, wr.to(torch.bfloat16).float(), br.to(torch.bfloat16).float())
+    torch.testing.assert_close(y.float().cpu(), yr, rtol=2e-2, atol=2e-2)
+    gy = torch.randn_like(y)
+    gx, gw, gb = torch.autograd.grad(y, [x, w, b], gy, create_graph=True)
+    gxr, gwr, gbr = torch.autograd.grad(yr, [xr, wr, br], gy.float().cpu(),
+                                        create_graph=True)
+    torch.testing.assert_close(gx.float().cpu(), gxr, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(gw.float().cpu(), gwr, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(gb.float().cpu(), gbr, rtol=2e-2, atol=2e-2)
+    # second order: grad of a grad-norm wrt the weight
+    s = (gx.float() ** 2).sum()
+    sw = torch.autograd.grad(s, w, retain_graph=True)[0]
+    sr = torch.autograd.grad((gxr ** 2).sum(), wr, retain_graph=True)[0]
+    torch.testing.assert_close(sw.cpu(), sr, rtol=5e-2, atol=5e-1)
+
+
+def test_linear_deterministic_across_fresh_calls():
+    torch.manual_seed(22)
+    outs = []
+    for trial in range(3):
+        x = torch.randn(2, 9, 64, device=dev()).to(torch.bfloat16)
+        torch.manual_seed(22)
+        x = torch.randn(2, 9, 64, device=dev()).to(torch.bfloat16)
+        w = torch.randn(2, 5, 64, device=dev())
+        outs.append(ops.task_linear(x, w, None).cpu())
+    assert (outs[0] == outs[1]).all().item() and (outs[1] == outs[2]).all().item()
