@@ -591,60 +591,45 @@ __global__ void repack_v2_kernel(const float* __restrict__ w,
 //     first layer's C in {1,3}, which v1 staged scalar).
 // ---------------------------------------------------------------------------
 
-// dY [T,NB,Ho,Wo,F] -> dYT [T,F,Kr]; block per (t, 256-wide k-tile):
-// reads are f-contiguous 16B chunks, writes are 512-byte contiguous runs
-// per f row (the per-(n,ho)-line version wrote <=96B bursts and was 13%
-// of step time).
-#define DYT_KT 256
+// dY [T,NB,Ho,Wo,F] -> dYT [T,F,Kr]; one block per (t,n,ho) line, tiled
+// in 64-wide w-chunks (any Wo).
 __global__ void dyt_pad_kernel(const bf16* __restrict__ dy,
                                bf16* __restrict__ dyt,
                                int T, int NB, int Ho, int Wo, int F,
                                int Wo8, long Kr) {
-  __shared__ short lds[DYT_KT * 72];  // [k_local][f] tile, stride 72
-  __shared__ long ksrc[DYT_KT];       // element offset into dy, or -1
+  __shared__ short lds[64 * 72];  // [f][wo] tile, stride 72 (16B-aligned)
   long b = blockIdx.x;
-  const long ntiles = (Kr + DYT_KT - 1) / DYT_KT;
-  const long kt = b % ntiles;
-  const int t = (int)(b / ntiles);
-  const long k0 = kt * DYT_KT;
-  // k -> (n, ho, wo) decode table (one division per k, 256 threads)
-  for (int kl = threadIdx.x; kl < DYT_KT; kl += blockDim.x) {
-    const long k = k0 + kl;
-    const int wo = (int)(k % Wo8);
-    const long line = k / Wo8;
-    const int ho = (int)(line % Ho);
-    const long n = line / Ho;
-    ksrc[kl] = (wo < Wo && n < NB)
-                   ? (((long)t * NB + n) * Ho + ho) * Wo * F + (long)wo * F
-                   : -1;
-  }
-  __syncthreads();
+  const int ho = (int)(b % Ho); b /= Ho;
+  const int n = (int)(b % NB); b /= NB;
+  const int t = (int)b;
   const int f8n = (F + 7) / 8;
-  for (int sid = threadIdx.x; sid < DYT_KT * f8n; sid += blockDim.x) {
-    const int kl = sid / f8n;
-    const int f0 = (sid % f8n) * 8;
-    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (ksrc[kl] >= 0) {
-      if (f0 + 8 <= F) {
-        v = *(const bf16x8*)&((const short*)dy)[ksrc[kl] + f0];
-      } else {
-        for (int j = 0; j < 8 && f0 + j < F; ++j)
-          v[j] = ((const short*)dy)[ksrc[kl] + f0 + j];
-      }
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) lds[kl * 72 + f0 + j] = v[j];
-  }
-  __syncthreads();
+  const short* src = (const short*)dy + ((((long)t * NB + n) * Ho + ho) * Wo) * F;
+  const long k0 = ((long)n * Ho + ho) * Wo8;
   short* dst = (short*)dyt + (long)t * F * Kr + k0;
-  const int kw = DYT_KT / 8;  // 16B writes per f row
-  for (int sid = threadIdx.x; sid < F * kw; sid += blockDim.x) {
-    const int f = sid / kw;
-    const int kl0 = (sid % kw) * 8;
-    bf16x8 v;
+  for (int wbase = 0; wbase < Wo8; wbase += 64) {
+    const int wlen = min(64, Wo8 - wbase);
+    for (int s = threadIdx.x; s < wlen * f8n; s += blockDim.x) {
+      const int wl = s / f8n;
+      const int wo = wbase + wl;
+      const int f0 = (s % f8n) * 8;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (wo < Wo && f0 + 8 <= F) {
+        v = *(const bf16x8*)&src[(long)wo * F + f0];
+      } else if (wo < Wo) {
+        for (int j = 0; j < 8 && f0 + j < F; ++j)
+          v[j] = src[(long)wo * F + f0 + j];
+      }
 #pragma unroll
-    for (int j = 0; j < 8; ++j) v[j] = lds[(kl0 + j) * 72 + f];
-    *(bf16x8*)&dst[(long)f * Kr + kl0] = v;
+      for (int j = 0; j < 8; ++j) lds[(f0 + j) * 72 + wl] = v[j];
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < F * (wlen / 8); s += blockDim.x) {
+      const int f = s / (wlen / 8);
+      const int w0 = (s % (wlen / 8)) * 8;
+      *(bf16x8*)&dst[(long)f * Kr + wbase + w0] =
+          *(const bf16x8*)&lds[f * 72 + w0];
+    }
+    __syncthreads();
   }
 }
 
@@ -661,64 +646,48 @@ __global__ void dyt_tail_kernel(bf16* __restrict__ dyt, int T, int F,
   }
 }
 
-// X [T,NB,H,W,C] -> XT [T,C,NB,Hxp,Wxp]; block per (t, n, 8-row batch,
-// 16-channel tile): writes become 8-row (~1.4 KB) contiguous bursts per
-// channel instead of single sub-128B rows.
-#define XT_HB 8
-#define XT_CT 16
+// X [T,NB,H,W,C] -> XT [T,C,NB,Hxp,Wxp]; block per (t, n, hxp, w-tile).
 __global__ void xt_pad_kernel(const bf16* __restrict__ x, bf16* __restrict__ xt,
                               int T, int NB, int H, int W, int C,
                               int Hxp, int Wxp, int pad) {
-  extern __shared__ short xlds[];  // [XT_HB][Wxp][XT_CT]
+  __shared__ short lds[64 * 72];  // [c][wx] tile
   long b = blockIdx.x;
-  const int ctiles = (C + XT_CT - 1) / XT_CT;
-  const int c0 = (int)(b % ctiles) * XT_CT; b /= ctiles;
-  const int hbn = (Hxp + XT_HB - 1) / XT_HB;
-  const int hb0 = (int)(b % hbn) * XT_HB; b /= hbn;
+  const int wt = (int)(b % ((Wxp + 63) / 64)); b /= (Wxp + 63) / 64;
+  const int hxp = (int)(b % Hxp); b /= Hxp;
   const int n = (int)(b % NB); b /= NB;
   const int t = (int)b;
-  const int cw = min(XT_CT, C - c0);
+  const int h = hxp - pad;
+  const int w0 = wt * 64;
+  const int c8n = (C + 7) / 8;
   const short* src = (const short*)x + (((long)t * NB + n) * H) * W * C;
-  // load phase: (row, wx, c-pair) slots; zero the pad border
-  const int rows = min(XT_HB, Hxp - hb0);
-  for (int sid = threadIdx.x; sid < rows * Wxp * (XT_CT / 8);
-       sid += blockDim.x) {
-    const int c8 = (sid % (XT_CT / 8)) * 8;
-    const int wx = (sid / (XT_CT / 8)) % Wxp;
-    const int r = sid / ((XT_CT / 8) * Wxp);
-    const int h = hb0 + r - pad;
-    const int w = wx - pad;
+  for (int s = threadIdx.x; s < 64 * c8n; s += blockDim.x) {
+    const int wl = s / c8n;
+    const int c0 = (s % c8n) * 8;
+    const int w = w0 + wl - pad;
     bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
     if (h >= 0 && h < H && w >= 0 && w < W) {
-      const long base = ((long)h * W + w) * C + c0 + c8;
-      if (c0 + c8 + 8 <= C) {
-        v = *(const bf16x8*)&src[base];
+      if (c0 + 8 <= C) {
+        v = *(const bf16x8*)&src[((long)h * W + w) * C + c0];
       } else {
-        for (int j = 0; j < 8 && c0 + c8 + j < C; ++j) v[j] = src[base + j];
+        for (int j = 0; j < 8 && c0 + j < C; ++j)
+          v[j] = src[((long)h * W + w) * C + c0 + j];
       }
     }
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      xlds[(r * Wxp + wx) * XT_CT + c8 + j] = v[j];
+    for (int j = 0; j < 8; ++j) lds[(c0 + j) * 72 + wl] = v[j];
   }
   __syncthreads();
-  // store phase: per (c, row) contiguous Wxp runs (8-row bursts per c)
-  short* dst = (short*)xt + (((long)t * C) * NB + n) * (long)Hxp * Wxp;
-  const int w8n = (Wxp + 7) / 8;
-  for (int sid = threadIdx.x; sid < cw * rows * w8n; sid += blockDim.x) {
-    const int w0 = (sid % w8n) * 8;
-    const int r = (sid / w8n) % rows;
-    const int c = c0 + sid / (w8n * rows);
-    const long o = ((long)c * NB * Hxp + hb0 + r) * Wxp + w0;
-    if (w0 + 8 <= Wxp) {
-      bf16x8 v;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        v[j] = xlds[(r * Wxp + w0 + j) * XT_CT + (c - c0)];
-      *(bf16x8*)&dst[o] = v;
+  short* dst = (short*)xt + ((((long)t * C) * NB + n) * Hxp + hxp) * Wxp;
+  for (int s = threadIdx.x; s < C * 8; s += blockDim.x) {
+    const int c = s / 8;
+    const int wg = (s % 8) * 8;
+    const int wx = w0 + wg;
+    if (wx + 8 <= Wxp) {
+      *(bf16x8*)&dst[((long)c * NB * Hxp) * Wxp + wx] =
+          *(const bf16x8*)&lds[c * 72 + wg];
     } else {
-      for (int j = 0; j < 8 && w0 + j < Wxp; ++j)
-        dst[o + j] = xlds[(r * Wxp + w0 + j) * XT_CT + (c - c0)];
+      for (int j = 0; j < 8 && wx + j < Wxp; ++j)
+        dst[((long)c * NB * Hxp) * Wxp + wx + j] = lds[c * 72 + wg + j];
     }
   }
 }
@@ -1348,9 +1317,7 @@ std::vector<torch::Tensor> tconv_wgrad_v2(torch::Tensor dy, torch::Tensor x,
   const int N9 = 9 * C;
   const int Wo8 = ((Wo + 7) / 8) * 8;
   const long Kprime = (long)NB * Ho * Wo8;
-  // round to the transpose k-tile (256) so every dyt column is written
-  // (zeros in the tail come from the transpose's own bounds decode)
-  const long Kr = ((Kprime + DYT_KT - 1) / DYT_KT) * DYT_KT;
+  const long Kr = ((Kprime + WBK - 1) / WBK) * WBK;
   const int Hxp = H + 2 * (int)pad;
   const int Wxp = Wo8 + 2;
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -1358,20 +1325,25 @@ std::vector<torch::Tensor> tconv_wgrad_v2(torch::Tensor dy, torch::Tensor x,
   // operand transposes
   auto dyt = torch::empty({T, F, Kr}, x.options());
   {
-    const long blocks = (long)T * (Kr / DYT_KT);
+    const long blocks = (long)T * NB * Ho;
     hipLaunchKernelGGL(dyt_pad_kernel, dim3((unsigned)blocks), dim3(256), 0,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(dyc.data_ptr()),
                        reinterpret_cast<bf16*>(dyt.data_ptr()),
                        T, NB, Ho, Wo, F, Wo8, Kr);
+    if (Kr > Kprime) {
+      const long total = (long)T * F * (Kr - Kprime);
+      hipLaunchKernelGGL(dyt_tail_kernel, dim3(ew_grid2(total, 256)), dim3(256),
+                         0, stream.stream(),
+                         reinterpret_cast<bf16*>(dyt.data_ptr()), T, F, Kprime,
+                         Kr);
+    }
   }
   auto xt = torch::empty({(long)T * C * NB * Hxp * Wxp}, x.options());
   {
-    const long blocks = (long)T * NB * ((Hxp + XT_HB - 1) / XT_HB) *
-                        ((C + XT_CT - 1) / XT_CT);
-    const int lds_bytes = XT_HB * Wxp * XT_CT * 2;
-    hipLaunchKernelGGL(xt_pad_kernel, dim3((unsigned)blocks), dim3(256),
-                       lds_bytes, stream.stream(),
+    const long blocks = (long)T * NB * Hxp * ((Wxp + 63) / 64);
+    hipLaunchKernelGGL(xt_pad_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                       stream.stream(),
                        reinterpret_cast<const bf16*>(xc.data_ptr()),
                        reinterpret_cast<bf16*>(xt.data_ptr()),
                        T, NB, H, W, C, Hxp, Wxp, (int)pad);
