@@ -377,7 +377,11 @@ class _ConvWgradFn(torch.autograd.Function):
                 and os.environ.get("MAML355_DCONV_WGRAD", "0") == "1"):
             dw, db = _ext().dconv_wgrad(dy, x, pad, with_bias)
         elif (dy.shape[4] <= 64
+                and dy.shape[1] * dy.shape[2] * dy.shape[3] >= 30000
                 and os.environ.get("MAML355_WGRAD_V2", "1") != "0"):
+            # v2's operand transposes are fixed cost: below ~30k reduction
+            # positions the v1 gather kernel wins (measured on the
+            # omniglot target passes: whole-bench +3% with v1 there)
             dw, db = _ext().tconv_wgrad_v2(dy, x, pad, with_bias)
         else:
             dw, db = _ext().tconv_wgrad(dy, x, pad, with_bias)
