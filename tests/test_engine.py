@@ -214,3 +214,26 @@ def test_fused_adam_clamp_matches_manual_clamp():
         opt1.step()
         opt2.step()
     torch.testing.assert_close(p1[0], p2[0], rtol=1e-5, atol=1e-6)
+
+
+def test_fused_adam_mixed_step_counts():
+    """Params with differing step counts (after a partial state load) are
+    bucketed and updated per their own bias correction."""
+    torch.manual_seed(3)
+    from howtotrainyourmamlpytorch_amd.meta.fused_adam import FusedAdam
+    p1 = torch.nn.Parameter(torch.randn(5))
+    p2 = torch.nn.Parameter(torch.randn(5))
+    opt = FusedAdam([p1, p2], lr=0.1)
+    p1.grad = torch.ones(5)
+    p2.grad = torch.ones(5)
+    opt.step()
+    # simulate a partial load: p2's state restarts
+    opt.state[p2]["step"] = torch.zeros(())
+    opt.state[p2]["exp_avg"].zero_()
+    opt.state[p2]["exp_avg_sq"].zero_()
+    p1.grad = torch.ones(5)
+    p2.grad = torch.ones(5)
+    opt.step()  # p1 at step 2, p2 at step 1 -> two buckets, no crash
+    ref = torch.nn.Parameter(torch.zeros(5))
+    assert opt.state[p1]["step"].item() == 2.0
+    assert opt.state[p2]["step"].item() == 1.0
