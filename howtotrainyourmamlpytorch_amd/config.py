@@ -156,6 +156,12 @@ def build_parser() -> argparse.ArgumentParser:
     # --- MI355X-native additions ---
     p.add_argument("--compute_dtype", type=str, default="bf16", choices=["bf16", "fp32"],
                    help="conv/linear compute dtype (fp32 accumulate either way)")
+    p.add_argument("--fp32_support_pass", type=str, default="False",
+                   help="run the inner-loop SUPPORT forward (and therefore its "
+                        "create_graph backward chain) in fp32 while target "
+                        "passes stay bf16 — tightens second-order "
+                        "meta-gradients at the cost of ATen convs on the "
+                        "support sets (which are small: N*S <= 100 images)")
     p.add_argument("--use_hip_kernels", type=str, default="True",
                    help="use the CDNA4 HIP extension on GPU (fail loudly if missing)")
     p.add_argument("--distributed_backend", type=str, default="auto",

@@ -147,7 +147,12 @@ class MAMLFewShotClassifier(nn.Module):
             self.device.type == "cuda"
             and getattr(self.args, "compute_dtype", "bf16") == "bf16"
         ) else torch.float32
-        xs = x_support.reshape(T, -1, *x_support.shape[-3:]).to(act_dtype)
+        # optional mixed-precision inner loop: support passes (and their
+        # create_graph backward chains, where second-order error
+        # accumulates) in fp32, target passes bf16
+        sup_dtype = torch.float32 if getattr(
+            self.args, "fp32_support_pass", False) else act_dtype
+        xs = x_support.reshape(T, -1, *x_support.shape[-3:]).to(sup_dtype)
         xt = x_target.reshape(T, -1, *x_target.shape[-3:]).to(act_dtype)
         ys = y_support.reshape(T, -1).long()
         yt = y_target.reshape(T, -1).long()
