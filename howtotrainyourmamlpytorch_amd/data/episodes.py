@@ -225,7 +225,10 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
         rng = np.random.RandomState(seed)
         classes = sorted(self.datasets[set_name].keys())
         chosen = rng.choice(len(classes), size=self.num_classes_per_set, replace=False)
-        k_per_class = rng.randint(0, 4, size=self.num_classes_per_set) if augment \
+        # per-class rot90 is the omniglot-style augmentation (reference
+        # data.py:92-95); cifar datasets augment by crop/flip instead
+        rotate = augment and "cifar" not in self.dataset_name.lower()
+        k_per_class = rng.randint(0, 4, size=self.num_classes_per_set) if rotate \
             else np.zeros(self.num_classes_per_set, dtype=np.int64)
         s, t = self.num_samples_per_class, self.num_target_samples
         images = np.zeros((self.num_classes_per_set, s + t, self.image_height,
@@ -241,6 +244,29 @@ class FewShotEpisodeDataset(torch.utils.data.Dataset):
                 images[ci, si] = img
         if self.image_channels == 3 and "imagenet" in self.dataset_name:
             images = (images - IMAGENET_MEAN) / IMAGENET_STD
+        elif "cifar" in self.dataset_name:
+            # reference transform set (data.py:81-90): train = RandomCrop
+            # 32/pad4 + RandomHorizontalFlip + Normalize(classification_
+            # mean/std); eval = Normalize only.  Crop/flip draws come from
+            # the episode RNG so episodes stay a pure function of the seed.
+            if augment:
+                h, w = self.image_height, self.image_width
+                for ci in range(images.shape[0]):
+                    for si in range(images.shape[1]):
+                        img = images[ci, si]
+                        padded = np.zeros((h + 8, w + 8, img.shape[2]),
+                                          dtype=img.dtype)
+                        padded[4:4 + h, 4:4 + w] = img
+                        oy, ox = rng.randint(0, 9), rng.randint(0, 9)
+                        img = padded[oy:oy + h, ox:ox + w]
+                        if rng.rand() < 0.5:
+                            img = img[:, ::-1]
+                        images[ci, si] = img
+            mean = np.asarray(getattr(self.args, "classification_mean",
+                                      [0.0, 0.0, 0.0]), dtype=np.float32)
+            std = np.asarray(getattr(self.args, "classification_std",
+                                     [1.0, 1.0, 1.0]), dtype=np.float32)
+            images = (images - mean) / std
         # HWC -> CHW
         x = torch.from_numpy(images).permute(0, 1, 4, 2, 3).contiguous()
         labels = torch.arange(self.num_classes_per_set).view(-1, 1).expand(

@@ -229,3 +229,32 @@ def test_shipped_label_maps_match_npz():
     names = [str(n) for n in data["class_names"]]
     assert len(fwd) == len(names) == 1623
     assert all(fwd[n] == i for i, n in enumerate(names))
+
+
+def test_cifar_transform_branch(tmp_path):
+    """cifar datasets use crop/flip + classification mean/std normalize
+    (reference data.py:81-90) and NOT the omniglot rotations."""
+    from PIL import Image
+    rng = np.random.RandomState(3)
+    root = tmp_path / "cifar10_fs"
+    for c in range(6):
+        d = root / "group" / f"cls{c}"
+        d.mkdir(parents=True)
+        for i in range(5):
+            arr = rng.randint(0, 255, size=(32, 32, 3), dtype=np.uint8)
+            Image.fromarray(arr, mode="RGB").save(d / f"{i}.png")
+    args = data_args(str(root), dataset_name="cifar10_fs")
+    args.image_height = 32
+    args.image_width = 32
+    args.image_channels = 3
+    args.classification_mean = [0.49, 0.48, 0.45]
+    args.classification_std = [0.25, 0.24, 0.26]
+    args.train_val_test_split = [0.5, 0.25, 0.25]
+    ds = FewShotEpisodeDataset(args, current_set="train")
+    xs_a, xt_a, *_ = ds.get_set("train", seed=5, augment=True)
+    xs_b, xt_b, *_ = ds.get_set("train", seed=5, augment=True)
+    torch.testing.assert_close(xs_a, xs_b)  # augment draws are seed-pure
+    xs_n, *_ = ds.get_set("train", seed=5, augment=False)
+    assert not torch.allclose(xs_a, xs_n)   # crop/flip changed pixels
+    # normalized: values centered near 0 with the given std
+    assert float(xs_n.mean().abs()) < 1.0 and float(xs_n.abs().max()) < 4.0
