@@ -61,3 +61,73 @@ def save_to_json(filename: str, dict_to_store: Dict[str, Any]) -> None:
 def load_from_json(filename: str) -> Dict[str, Any]:
     with open(filename, "r") as f:
         return json.load(f)
+
+
+# ---------------------------------------------------------------------------
+# experiment-log JSON helpers (reference: ``utils/storage.py:82-128`` —
+# defined there but never called by the reference's own builder; kept for
+# API-surface parity and usable programmatically).
+# ---------------------------------------------------------------------------
+def create_json_experiment_log(experiment_log_dir: str, args,
+                               log_name: str = "experiment_log.json") -> str:
+    import datetime
+    path = os.path.join(experiment_log_dir, log_name)
+    summary = {k: v for k, v in vars(args).items()
+               if isinstance(v, (int, float, str, bool, list, type(None)))}
+    ts = datetime.datetime.now().timestamp()
+    summary["epoch_stats"] = {}
+    summary["experiment_status"] = [(ts, "initialization")]
+    summary["experiment_initialization_time"] = ts
+    with open(path, "w") as f:
+        json.dump(summary, f)
+    return path
+
+
+def update_json_experiment_log_dict(key: str, value, experiment_log_dir: str,
+                                    log_name: str = "experiment_log.json") -> None:
+    path = os.path.join(experiment_log_dir, log_name)
+    with open(path) as f:
+        summary = json.load(f)
+    summary[key].append(value)
+    with open(path, "w") as f:
+        json.dump(summary, f)
+
+
+def change_json_log_experiment_status(experiment_status: str,
+                                      experiment_log_dir: str,
+                                      log_name: str = "experiment_log.json") -> None:
+    import datetime
+    update_json_experiment_log_dict(
+        "experiment_status",
+        (datetime.datetime.now().timestamp(), experiment_status),
+        experiment_log_dir, log_name)
+
+
+def update_json_experiment_log_epoch_stats(epoch_stats: Dict[str, Any],
+                                           experiment_log_dir: str,
+                                           log_name: str = "experiment_log.json") -> str:
+    path = os.path.join(experiment_log_dir, log_name)
+    with open(path) as f:
+        summary = json.load(f)
+    stats = summary["epoch_stats"]
+    for k, v in epoch_stats.items():
+        stats.setdefault(k, []).append(float(v))
+    with open(path, "w") as f:
+        json.dump(summary, f)
+    return path
+
+
+def get_best_validation_model_statistics(log_dir: str,
+                                         filename: str = "summary_statistics.csv",
+                                         key: str = "val_accuracy_mean"):
+    """Best validation value and its epoch from the stats CSV (reference:
+    ``utils/storage.py:68-80``; generalized to maximize accuracy keys and
+    minimize loss keys)."""
+    import numpy as np
+    stats = load_statistics(log_dir, filename)
+    vals = np.array([float(v) for v in stats[key]], dtype=np.float64)
+    if "loss" in key:
+        idx = int(np.argmin(vals))
+    else:
+        idx = int(np.argmax(vals))
+    return float(vals[idx]), idx

@@ -149,3 +149,26 @@ def test_chunked_train_step_losses_are_weighted_means():
     # MSL importance entries survive chunking and sum to ~1
     iv = [v for k, v in losses.items() if k.startswith("loss_importance_vector_")]
     assert iv and abs(sum(iv) - 1.0) < 1e-6
+
+
+def test_experiment_log_json_helpers(tmp_path):
+    from howtotrainyourmamlpytorch_amd.experiment import storage as st
+    from howtotrainyourmamlpytorch_amd.config import get_args
+    args = get_args(["--experiment_name", "x"])
+    p = st.create_json_experiment_log(str(tmp_path), args)
+    st.change_json_log_experiment_status("training", str(tmp_path))
+    st.update_json_experiment_log_epoch_stats(
+        {"val_accuracy_mean": 0.5, "val_loss_mean": 1.2}, str(tmp_path))
+    st.update_json_experiment_log_epoch_stats(
+        {"val_accuracy_mean": 0.8, "val_loss_mean": 0.7}, str(tmp_path))
+    import json as js
+    d = js.load(open(p))
+    assert d["epoch_stats"]["val_accuracy_mean"] == [0.5, 0.8]
+    assert d["experiment_status"][-1][1] == "training"
+    # best-val helper over a real CSV
+    st.save_statistics(str(tmp_path), ["epoch", "val_accuracy_mean"], create=True)
+    st.save_statistics(str(tmp_path), [0, 0.4])
+    st.save_statistics(str(tmp_path), [1, 0.9])
+    st.save_statistics(str(tmp_path), [2, 0.6])
+    best, epoch = st.get_best_validation_model_statistics(str(tmp_path))
+    assert best == 0.9 and epoch == 1
