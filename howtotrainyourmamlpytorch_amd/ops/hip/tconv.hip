@@ -10,20 +10,26 @@
 //   fwd   : Y[t]  = im2col(X[t])  @ Wp[t]        M=NS*Ho*Wo, N=Cout, K=9*Cin
 //   dgrad : dX[t] = im2col(dY[t]) @ Wp_flip[t]   (SAME kernel, weights
 //           repacked flipped+transposed — full correlation identity)
-//   wgrad : dW[t] = dY[t]^T @ im2col(X[t])       split-K, fp32 atomics
+//   wgrad : dW[t] = dY[t]^T @ im2col(X[t])       split-K (v2: operand
+//           transposes make both sides k-contiguous for async staging)
 //
 // Data: bf16 activations/weights (repacked from the fp32 arena), fp32
 // accumulate via v_mfma_f32_16x16x32_bf16.  Layout: NHWC (channel-
 // innermost = K-contiguous im2col rows).
 //
-// Geometry (fwd/dgrad): block = 512 threads = 8 waves; BM=128 output rows,
-// BN = Cout (<= 64, i.e. up to 4 n-tiles of 16), BK=64 (2 MFMA K-slices
-// per barrier).  Wave w owns 16-row m-subtile w and iterates all n-tiles.
-// A-tile staged by im2col gather, B-tile contiguous from the repacked
-// weights, transposed into LDS for K-contiguous fragment reads; both
-// tiles XOR-swizzled (swz64).  Geometry settled by same-box A/B sweeps —
-// see profiles/README.md (BM 64->128 +50% TF; BK=128 and 2-half-wave
-// variants measured slower via LDS-occupancy / issue-pressure cliffs).
+// Two fwd/dgrad generations, both BM=256 / 8 waves / BK=64, bitwise-equal
+// outputs (tests pin this):
+//   v1: synchronous register staging between two barriers (im2col gather
+//       + transposed B writes) — the fallback for Ci % 8 != 0.
+//   v2 (default, single-buffer): `global_load_lds` async DMA staging with
+//       a padded input (no boundary predicates), a pre-swizzled weight
+//       LDS image, per-thread incremental k-state and a zero page for the
+//       tails.  Measured: conv1 192->222 TF, omiglot conv1 289->312; the
+//       double-buffered 2-phase variant (MAML355_CONV_V2_SBUF=0) LOSES
+//       through occupancy (80 KB LDS -> 2 blocks/CU).
+// All geometry settled by same-box A/B sweeps — profiles/README.md
+// (r1: BM 64->128->256; BK=128 and 2-half-wave variants measured slower
+// via LDS-occupancy / issue-pressure cliffs; r2: the pipeline ladder).
 
 #include "common.h"
 #include <torch/extension.h>
